@@ -1,0 +1,6 @@
+#!/usr/bin/env python3
+"""Compatibility entry point (reference: ``src/AdHoc_train.py``)."""
+from multihop_offload_amd.harness.adhoc_train import main
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,236 @@
+"""Single-case environment: offloading decisions, greedy routing, analytic
+queueing evaluation.
+
+Clean-room reimplementation of the decision/evaluation half of ``AdhocCloud``
+(``/root/reference/src/offloading_v3.py:341-550``) on top of ``CaseGraph``.
+This is the *oracle* path (numpy, single graph) used by the compatible
+harnesses and as the ground truth for the batched device engine
+(``engine.py``) and the HIP kernels (``ops/``).
+
+Semantics preserved exactly (verified by tests against hand-derived cases):
+  * baseline distance matrix: 1/link_rate off-diagonal, 1/proc_bw diagonal
+    (``offloading_v3.py:341-361``)
+  * offloading cost: max(sp_ul*ul, hops) + max(sp_dl*dl, hops) +
+    max(proc*ul, 1), explore / argmin / softmax-sample
+    (``offloading_v3.py:388-439``; the softmax prob mode reproduces the
+    reference's high-cost-preferring behavior behind ``prob=True``)
+  * greedy next-hop routing: argmin over neighbors of sp[nb, dst], ascending
+    node-id tie-break (``offloading_v3.py:441-453``)
+  * empirical evaluation with per-job congestion fallbacks
+    (``offloading_v3.py:455-550``)
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import List, Optional
+
+import numpy as np
+
+from .graphs import CaseGraph, JobInstance
+
+
+@dataclasses.dataclass
+class Flow:
+    """Reference ``Flow`` (``offloading_v3.py:140-150``), route as node list."""
+    src: int
+    dst: int
+    route: List[int]
+    nhop: int
+
+
+def softmax(x: np.ndarray) -> np.ndarray:
+    """Reference ``util.softmax`` (``util.py:113-116``) — NOT max-subtracted,
+    NOT negated: prefers high-cost entries. Kept verbatim for parity."""
+    ex = np.exp(x)
+    return ex / ex.sum()
+
+
+def apsp(g: CaseGraph, link_weights: Optional[np.ndarray] = None) -> np.ndarray:
+    """All-pairs shortest path distances over the connectivity graph.
+    ``link_weights=None`` → hop counts. Reference: ``util.py:101-110``
+    (networkx Dijkstra); here scipy's C Dijkstra on the canonical CSR."""
+    from scipy.sparse import csr_matrix
+    from scipy.sparse.csgraph import shortest_path
+    if link_weights is None:
+        return g.sp_hop
+    N = g.num_nodes
+    w = np.zeros((N, N))
+    w[g.edges[:, 0], g.edges[:, 1]] = link_weights
+    w[g.edges[:, 1], g.edges[:, 0]] = link_weights
+    mask = g.adj != 0
+    data = csr_matrix((w[mask], np.nonzero(mask)), shape=(N, N))
+    return shortest_path(data, method="D", directed=False)
+
+
+class AdhocCloudEnv:
+    """Decision + evaluation wrapper around one ``CaseGraph`` and one
+    ``JobInstance``."""
+
+    def __init__(self, g: CaseGraph):
+        self.g = g
+        self.jobs: Optional[JobInstance] = None
+        self.flows: List[Flow] = []
+
+    # -- job management (offloading_v3.py:243-250) ----------------------------
+    def set_jobs(self, jobs: JobInstance):
+        self.jobs = jobs
+        self.flows = []
+
+    # -- baseline distance matrix (offloading_v3.py:341-361) ------------------
+    def dmtx_baseline(self):
+        g = self.g
+        N = g.num_nodes
+        dmtx = np.full((N, N), np.inf)
+        with np.errstate(divide="ignore"):
+            dproc = 1.0 / g.proc_bws            # inf at relays (bw=0)
+            dlist = 1.0 / g.link_rates
+        np.fill_diagonal(dmtx, dproc)
+        dmtx[g.edges[:, 0], g.edges[:, 1]] = dlist
+        dmtx[g.edges[:, 1], g.edges[:, 0]] = dlist
+        return dmtx, dlist, dproc
+
+    # -- local computing (offloading_v3.py:363-386) ---------------------------
+    def local_compute(self, unit_delay_servers: np.ndarray):
+        jobs = self.jobs
+        self.flows = [Flow(int(s), int(s), [int(s), int(s)], 0)
+                      for s in jobs.sources]
+        delays = np.maximum(unit_delay_servers[jobs.sources] * jobs.ul, 1.0)
+        return jobs.sources.copy(), delays
+
+    # -- offloading decision (offloading_v3.py:388-439) -----------------------
+    def offloading(self, spmtx_in: np.ndarray, hpmtx: np.ndarray,
+                   explore: float = 0.0, prob: bool = False,
+                   rng: Optional[np.random.RandomState] = None):
+        rng = rng or np.random
+        g, jobs = self.g, self.jobs
+        servers = np.asarray(g.servers, dtype=np.int64)
+        S = len(servers)
+        unit_delay_servers = np.diagonal(spmtx_in)
+        spmtx = spmtx_in.copy()
+        np.fill_diagonal(spmtx, 0)
+
+        src = jobs.sources
+        local_delay = unit_delay_servers[src] * jobs.ul                    # (J,)
+        ul_d = np.maximum(spmtx[src][:, servers] * jobs.ul[:, None],
+                          hpmtx[src][:, servers])                          # (J,S)
+        dl_d = np.maximum(spmtx[servers][:, src].T * jobs.dl[:, None],
+                          hpmtx[servers][:, src].T)
+        proc_d = np.maximum(unit_delay_servers[servers][None, :] * jobs.ul[:, None],
+                            1.0)
+        server_delays = ul_d + dl_d + proc_d                               # (J,S)
+        costs = np.concatenate([server_delays, local_delay[:, None]], axis=1)
+
+        decisions, delays = [], []
+        self.flows = []
+        for j in range(jobs.num_jobs):
+            if rng.uniform(0, 1) < explore:
+                jidx = rng.choice(S + 1)
+            elif not prob:
+                jidx = int(np.argmin(costs[j]))
+            else:
+                jidx = rng.choice(S + 1, p=softmax(costs[j]))
+            s = int(src[j])
+            if jidx < S:
+                dst = int(servers[jidx])
+                route, nhop = self.routing(s, dst, spmtx)
+                job_delay = server_delays[j, jidx]
+            else:
+                dst, route, nhop = s, [s, s], 0
+                job_delay = local_delay[j]
+            self.flows.append(Flow(s, dst, route, nhop))
+            decisions.append(dst)
+            delays.append(job_delay)
+        return np.asarray(decisions), np.asarray(delays)
+
+    # -- greedy next-hop walk (offloading_v3.py:441-453) ----------------------
+    def routing(self, src: int, dst: int, spmtx: np.ndarray):
+        g = self.g
+        route = [src]
+        node, nhop = src, 0
+        while node != dst:
+            lo, hi = g.adj_indptr[node], g.adj_indptr[node + 1]
+            nbs = g.adj_indices[lo:hi]
+            node = int(nbs[np.argmin(spmtx[nbs, dst])])
+            nhop += 1
+            route.append(node)
+            if nhop > g.num_nodes:
+                raise RuntimeError("greedy route failed to terminate")
+        return route, nhop
+
+    # -- route → link-id sequences --------------------------------------------
+    def route_links(self, flow: Flow) -> np.ndarray:
+        """Link ids along a route (empty for local computing)."""
+        g = self.g
+        if flow.src == flow.dst:
+            return np.empty(0, dtype=np.int64)
+        r = np.asarray(flow.route)
+        return g.link_matrix[r[:-1], r[1:]]
+
+    # -- analytic queueing evaluation (offloading_v3.py:455-550) --------------
+    def run(self):
+        g, jobs = self.g, self.jobs
+        N, E, J = g.num_nodes, g.num_links, jobs.num_jobs
+        assert len(self.flows) == J
+        server_delay = np.full((N, J), np.nan)
+        link_delay = np.full((E, J), np.nan)
+        link_load = np.zeros((E, J))
+        server_load = np.zeros(N)
+
+        ul_rate = jobs.ul * jobs.rates
+        dl_rate = jobs.dl * jobs.rates
+        for j, flow in enumerate(self.flows):
+            links = self.route_links(flow)
+            np.add.at(link_load, (links, np.full(len(links), j)),
+                      ul_rate[j] + dl_rate[j])
+            server_load[flow.dst] += ul_rate[j]
+
+        link_lambda = link_load.sum(axis=1)
+        link_mu = g.link_rates / (g.cf_degs + 1.0)
+        for _ in range(10):
+            busy = np.clip(link_lambda / link_mu, 0, 1.0)
+            nb = np.zeros(E)
+            np.add.at(nb, _conf_rows(g), busy[g.conf_indices])
+            link_mu = g.link_rates / (1.0 + nb)
+
+        unit_mtx = np.full((N, N), np.nan)
+        for j, flow in enumerate(self.flows):
+            job_tot = jobs.ul[j] + jobs.dl[j]
+            nhop = float(flow.nhop)
+            for lidx in self.route_links(flow):
+                gap = link_mu[lidx] - link_lambda[lidx]
+                if gap <= 0:
+                    unit = float(g.T) * link_lambda[lidx] / (job_tot * link_mu[lidx])
+                else:
+                    unit = 1.0 / gap
+                u, v = g.edges[lidx]
+                unit_mtx[u, v] = unit_mtx[v, u] = unit
+                link_delay[lidx, j] = (max(jobs.ul[j] * unit, nhop)
+                                       + max(jobs.dl[j] * unit, nhop))
+            dst = flow.dst
+            gap = g.proc_bws[dst] - server_load[dst]
+            if gap <= 0:
+                unit = float(g.T) * server_load[dst] / (jobs.ul[j] * g.proc_bws[dst])
+            else:
+                unit = 1.0 / gap
+            unit_mtx[dst, dst] = unit
+            server_delay[dst, j] = max(jobs.ul[j] * unit, 1.0)
+        return link_delay, server_delay, unit_mtx
+
+
+_CONF_ROWS_CACHE = {}
+
+
+def _conf_rows(g: CaseGraph) -> np.ndarray:
+    key = id(g)
+    got = _CONF_ROWS_CACHE.get(key)
+    if got is None or len(got) != len(g.conf_indices):
+        got = np.repeat(np.arange(g.num_links), np.diff(g.conf_indptr))
+        _CONF_ROWS_CACHE[key] = got
+    return got
+
+
+def delay_empirical(link_delay: np.ndarray, server_delay: np.ndarray) -> np.ndarray:
+    """Per-job total delay: nansum over links + nansum over servers
+    (``AdHoc_train.py:140,153``)."""
+    return np.nansum(link_delay, axis=0) + np.nansum(server_delay, axis=0)

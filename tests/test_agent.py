@@ -1,0 +1,222 @@
+"""Agent: critic gradients vs finite differences, route-bias VJP vs a tape
+reimplementation, checkpoint roundtrip, replay/constraint semantics."""
+import numpy as np
+import pytest
+import torch
+
+from multihop_offload_amd import ACOAgent, AdhocCloudEnv, JobInstance
+from multihop_offload_amd.agent import AgentConfig
+from multihop_offload_amd.queueing import (ConflictCSR, delay_with_fallback,
+                                           fixed_point_mu)
+
+
+def _agent(**kw):
+    cfg = AgentConfig(T=1000, seed=123, **kw)
+    return ACOAgent(cfg, 100)
+
+
+def _env(small_case, load=0.15, seed=1):
+    rng = np.random.RandomState(seed)
+    jobs = JobInstance.sample(small_case.mobile_nodes, load, rng)
+    env = AdhocCloudEnv(small_case)
+    env.set_jobs(jobs)
+    return env
+
+
+def test_forward_shapes_and_nonneg(small_case):
+    agent = _agent()
+    env = _env(small_case)
+    dm, ld, nd = agent.forward(small_case, env.jobs)
+    N = small_case.num_nodes
+    assert dm.shape == (N, N)
+    # model output is ReLU'd → λ ≥ 0 → delays are 1/(mu-λ) with λ finite
+    assert torch.isfinite(ld).all()
+
+
+def test_critic_grad_routes_matches_finite_differences(small_case):
+    """∂loss/∂routes (the analog of gnn_offloading_agent.py:374) via torch
+    autograd must match central finite differences of the critic loss."""
+    g = small_case
+    env = _env(g)
+    agent = _agent()
+    agent.forward_backward(env, 0.0, np.random.RandomState(2))
+    ext = g.ext
+    E, Ee, J = g.num_links, ext.num_edges_ext, env.jobs.num_jobs
+    jobs = env.jobs
+
+    routes_np = np.zeros((Ee, J))
+    for j, flow in enumerate(env.flows):
+        links = env.route_links(flow)
+        seq = np.concatenate([links, [ext.node_vedge[flow.dst]]])
+        routes_np[seq, j] = 1.0
+
+    conf = ConflictCSR(g.conf_indptr, g.conf_indices)
+    rates = torch.tensor(g.link_rates)
+    cf = torch.tensor(g.cf_degs)
+    bw = torch.tensor(g.proc_bws[ext.comp_nodes])
+    jl = torch.tensor((jobs.rates * jobs.ul)[:, None])
+    jd = torch.tensor((jobs.ul + jobs.dl)[None, :])
+
+    def loss_of(routes_t):
+        ll = (routes_t @ jl)[:, 0]
+        mu = fixed_point_mu(ll[:E], rates, cf, conf)
+        l_d = delay_with_fallback(ll[:E], mu, g.T, 101.0)
+        n_d = delay_with_fallback(ll[E:], bw, g.T, 100.0)
+        unit = torch.cat([l_d, n_d])
+        return torch.maximum(jd * unit[:, None] * routes_t, routes_t).sum()
+
+    routes = torch.tensor(routes_np, requires_grad=True)
+    loss = loss_of(routes)
+    (grad,) = torch.autograd.grad(loss, routes)
+
+    rng = np.random.RandomState(0)
+    eps = 1e-6
+    for _ in range(12):
+        e = rng.randint(Ee)
+        j = rng.randint(J)
+        if routes_np[e, j] == 0:
+            continue     # FD at the max() tie of zero-entries is undefined
+        rp, rm = routes_np.copy(), routes_np.copy()
+        rp[e, j] += eps
+        rm[e, j] -= eps
+        fd = (loss_of(torch.tensor(rp)) - loss_of(torch.tensor(rm))).item() / (2 * eps)
+        assert np.isclose(grad[e, j].item(), fd, rtol=1e-4, atol=1e-5)
+
+
+def test_route_bias_scan_matches_tape_semantics(small_case):
+    """The closed-form prefix-scan route-bias VJP must equal a literal
+    autograd transcription of the reference tape (gnn_offloading_agent.py:
+    384-409): bias[e,j] = suffix sum of unit delays, cotangent -grad_routes."""
+    g = small_case
+    env = _env(g, seed=7)
+    agent = _agent()
+    agent.forward_backward(env, 0.0, np.random.RandomState(3))
+    ext = g.ext
+    Ee, J = ext.num_edges_ext, env.jobs.num_jobs
+
+    route_seqs = []
+    for j, flow in enumerate(env.flows):
+        links = env.route_links(flow)
+        route_seqs.append(np.concatenate([links, [ext.node_vedge[flow.dst]]]))
+
+    rng = np.random.RandomState(4)
+    grad_routes = rng.normal(size=(Ee, J))
+    unit = torch.tensor(rng.uniform(0.1, 2.0, Ee), requires_grad=True)
+
+    # tape transcription: build bias via scatter, backprop -grad_routes
+    bias_cols = []
+    for j, seq in enumerate(route_seqs):
+        col = torch.zeros(Ee, dtype=unit.dtype)
+        tmp = torch.zeros((), dtype=unit.dtype)
+        for e in reversed(seq):
+            val = unit[e] + tmp
+            col = col.index_put((torch.tensor([e]),), val.reshape(1))
+            tmp = val
+        bias_cols.append(col)
+    bias = torch.stack(bias_cols, dim=1)
+    (want,) = torch.autograd.grad(
+        bias, unit, grad_outputs=torch.tensor(-grad_routes))
+
+    # closed form (what agent.forward_backward uses)
+    got = np.zeros(Ee)
+    for j, seq in enumerate(route_seqs):
+        np.add.at(got, seq, -np.cumsum(grad_routes[seq, j]))
+    assert np.allclose(got, want.numpy(), rtol=1e-10)
+
+
+def test_actor_vjp_matches_finite_differences(small_case):
+    """g.gradient(delay_mtx, weights, output_gradients=G) equivalence: the
+    returned parameter gradient must equal d(sum(G⊙dm))/dθ by FD."""
+    g = small_case
+    env = _env(g)
+    agent = _agent()
+    rng = np.random.RandomState(5)
+    G = rng.normal(size=(g.num_nodes, g.num_nodes))
+    Gt = torch.tensor(G)
+
+    dm, _, _ = agent.forward(g, env.jobs)
+    params = list(agent.model.parameters())
+    grads = torch.autograd.grad(dm, params, grad_outputs=Gt)
+
+    def scalar():
+        with torch.no_grad():
+            pass
+        dm2, _, _ = agent.forward(g, env.jobs)
+        # mask the inf diagonal (relays): constant → no gradient anyway
+        m = torch.isfinite(dm2)
+        return (dm2[m] * Gt[m]).sum()
+
+    p = params[0]
+    eps = 1e-6
+    for idx in [(0, 0, 0), (0, 2, 7), (1, 3, 15)]:
+        with torch.no_grad():
+            old = p[idx].item()
+            p[idx] = old + eps
+            fp = scalar().item()
+            p[idx] = old - eps
+            fm = scalar().item()
+            p[idx] = old
+        fd = (fp - fm) / (2 * eps)
+        assert np.isclose(grads[0][idx].item(), fd, rtol=1e-4, atol=1e-6)
+
+
+def test_forward_backward_produces_finite_grads(small_case):
+    env = _env(small_case)
+    agent = _agent()
+    # avoid an (init-dependent) dead output ReLU: λ≡0 would make the actor
+    # gradient exactly zero, which is correct but vacuous for this test —
+    # shrink the weights so the final pre-activation ≈ bias > 0
+    with torch.no_grad():
+        for layer in agent.model.layers:
+            layer.weight.mul_(0.01)
+        agent.model.layers[-1].bias.fill_(0.5)
+    agent.forward_backward(env, 0.0, np.random.RandomState(0))
+    grads, loss, mse = agent.memory[-1]
+    assert all(torch.isfinite(gr).all() for gr in grads)
+    assert np.isfinite(loss) and np.isfinite(mse)
+    assert any(gr.abs().sum() > 0 for gr in grads)
+
+
+def test_replay_and_constraints(small_case):
+    env = _env(small_case)
+    agent = _agent()
+    with torch.no_grad():
+        for layer in agent.model.layers:
+            layer.weight.mul_(0.01)
+        agent.model.layers[-1].bias.fill_(0.5)
+    assert np.isnan(agent.replay(10))          # not enough memory
+    for i in range(12):
+        agent.forward_backward(env, 0.0, np.random.RandomState(i))
+    before = [p.clone() for p in agent.model.parameters()]
+    loss = agent.replay(10)
+    assert np.isfinite(loss)
+    after = list(agent.model.parameters())
+    assert any(not torch.equal(b, a) for b, a in zip(before, after))
+    # max_norm constraints hold
+    for layer in agent.model.layers:
+        norms = torch.sqrt((layer.weight ** 2).sum(dim=0))
+        assert (norms <= 1.0 + 1e-9).all()
+        assert layer.bias.norm() <= 1.0 + 1e-9
+    # epsilon decayed exactly once
+    assert np.isclose(agent.epsilon, 1.0 * agent.cfg.epsilon_decay)
+
+
+def test_checkpoint_roundtrip(tmp_path, small_case):
+    agent = _agent()
+    path = str(tmp_path / "model_ChebConv_X_a5_c5_ACO_agent")
+    agent.save(path + "/cp-0000.ckpt")
+    agent2 = _agent()
+    # different init (no seed sharing of state) → load → identical
+    agent2.load(path)
+    for p1, p2 in zip(agent.model.parameters(), agent2.model.parameters()):
+        assert torch.equal(p1, p2)
+    import os
+    assert os.path.isfile(path + "/checkpoint")
+    assert os.path.isfile(path + "/cp-0000.ckpt.npz")
+
+
+def test_param_count_matches_reference_at_k1():
+    # shipped reference model: 3,361 fp64 params
+    agent = _agent(K=1)
+    n = sum(p.numel() for p in agent.model.parameters())
+    assert n == 3361

@@ -1,0 +1,208 @@
+"""Environment vs literal per-job transcriptions of the reference semantics
+(offloading_v3.py:341-550) and networkx APSP oracles."""
+import numpy as np
+import networkx as nx
+import pytest
+
+from multihop_offload_amd import AdhocCloudEnv, CaseGraph, JobInstance
+from multihop_offload_amd.env import apsp, delay_empirical, softmax
+
+
+def _nx_graph(g, link_weights=None):
+    G = nx.from_numpy_array(g.adj.astype(float))
+    if link_weights is not None:
+        for l, (u, v) in enumerate(g.edges):
+            G[u][v]["delay"] = link_weights[l]
+    return G
+
+
+def test_apsp_vs_networkx(small_case):
+    g = small_case
+    rng = np.random.RandomState(0)
+    w = rng.uniform(0.01, 2.0, g.num_links)
+    got = apsp(g, w)
+    G = _nx_graph(g, w)
+    lengths = dict(nx.all_pairs_dijkstra_path_length(G, weight="delay"))
+    want = np.zeros_like(got)
+    for i in range(g.num_nodes):
+        for j in range(g.num_nodes):
+            want[i, j] = lengths[i][j]
+    assert np.allclose(got, want)
+    hops = apsp(g)
+    lengths = dict(nx.all_pairs_dijkstra_path_length(G, weight=None))
+    for i in range(g.num_nodes):
+        for j in range(g.num_nodes):
+            assert hops[i, j] == lengths[i][j]
+
+
+def _reference_offloading(g, jobs, spmtx_in, hpmtx):
+    """Literal transcription of offloading_v3.py:388-439 (argmin mode)."""
+    servers = list(g.servers)
+    uds = np.diagonal(spmtx_in)
+    sp = spmtx_in.copy()
+    np.fill_diagonal(sp, 0)
+    decisions, delays, routes = [], [], []
+    for j in range(jobs.num_jobs):
+        src = int(jobs.sources[j])
+        ul, dl = jobs.ul[j], jobs.dl[j]
+        local = uds[src] * ul
+        s_ul = sp[src, servers] * ul
+        s_dl = sp[servers, src] * dl
+        s_pr = uds[servers] * ul
+        ul_d = np.max([s_ul, hpmtx[src, servers]], axis=0)
+        dl_d = np.max([s_dl, hpmtx[servers, src]], axis=0)
+        pr_d = np.max([s_pr, np.ones_like(s_pr)], axis=0)
+        costs = np.append(ul_d + dl_d + pr_d, local)
+        jidx = int(np.argmin(costs))
+        if jidx < len(servers):
+            dst = servers[jidx]
+            # greedy walk
+            route, node, nh = [src], src, 0
+            while node != dst:
+                nbs = np.nonzero(g.adj[node])[0]
+                node = int(nbs[np.argmin(sp[nbs, dst])])
+                nh += 1
+                route.append(node)
+            delays.append(costs[jidx])
+        else:
+            dst, route, nh = src, [src, src], 0
+            delays.append(local)
+        decisions.append(dst)
+        routes.append((route, nh))
+    return np.array(decisions), np.array(delays), routes
+
+
+def _reference_run(g, jobs, flows):
+    """Literal transcription of offloading_v3.py:455-550."""
+    N, E, J = g.num_nodes, g.num_links, jobs.num_jobs
+    link_load = np.zeros((E, J))
+    server_load = np.zeros(N)
+    for j in range(J):
+        ulr = jobs.ul[j] * jobs.rates[j]
+        dlr = jobs.dl[j] * jobs.rates[j]
+        f = flows[j]
+        if f.src != f.dst:
+            n0 = f.src
+            for n1 in f.route[1:]:
+                l = g.link_matrix[n0, n1]
+                link_load[l, j] += ulr + dlr
+                n0 = n1
+        server_load[f.dst] += ulr
+    lam = link_load.sum(axis=1)
+    mu = g.link_rates / (g.cf_degs + 1.0)
+    rows = np.repeat(np.arange(E), np.diff(g.conf_indptr))
+    for _ in range(10):
+        busy = np.clip(lam / mu, 0, 1.0)
+        nb = np.zeros(E)
+        np.add.at(nb, rows, busy[g.conf_indices])
+        mu = g.link_rates / (1.0 + nb)
+    ldel = np.full((E, J), np.nan)
+    sdel = np.full((N, J), np.nan)
+    umtx = np.full((N, N), np.nan)
+    for j in range(J):
+        f = flows[j]
+        tot = jobs.ul[j] + jobs.dl[j]
+        nh = float(f.nhop)
+        if f.src != f.dst:
+            n0 = f.src
+            for n1 in f.route[1:]:
+                l = g.link_matrix[n0, n1]
+                unit = (1.0 / (mu[l] - lam[l]) if mu[l] - lam[l] > 0
+                        else float(g.T) * lam[l] / (tot * mu[l]))
+                umtx[n0, n1] = umtx[n1, n0] = unit
+                ldel[l, j] = max(jobs.ul[j] * unit, nh) + max(jobs.dl[j] * unit, nh)
+                n0 = n1
+        dst = f.dst
+        gap = g.proc_bws[dst] - server_load[dst]
+        unit = (1.0 / gap if gap > 0
+                else float(g.T) * server_load[dst] / (jobs.ul[j] * g.proc_bws[dst]))
+        umtx[dst, dst] = unit
+        sdel[dst, j] = max(jobs.ul[j] * unit, 1.0)
+    return ldel, sdel, umtx
+
+
+@pytest.mark.parametrize("load", [0.15, 0.6])
+def test_offloading_and_run_vs_reference_semantics(small_case, load):
+    g = small_case
+    rng = np.random.RandomState(9)
+    jobs = JobInstance.sample(g.mobile_nodes, load, rng)
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    dmtx, dlist, dproc = env.dmtx_baseline()
+    dproc2 = np.where(dproc > 0, dproc, float(g.T))
+    sp = apsp(g, dlist)
+    np.fill_diagonal(sp, dproc2)
+    hp = g.sp_hop
+
+    dec, dly = env.offloading(sp, hp)
+    dec_ref, dly_ref, routes_ref = _reference_offloading(g, jobs, sp, hp)
+    assert np.array_equal(dec, dec_ref)
+    assert np.allclose(dly, dly_ref)
+    for f, (route, nh) in zip(env.flows, routes_ref):
+        assert f.route == route and f.nhop == nh
+
+    ldel, sdel, umtx = env.run()
+    ldel_r, sdel_r, umtx_r = _reference_run(g, jobs, env.flows)
+    assert np.allclose(ldel, ldel_r, equal_nan=True)
+    assert np.allclose(sdel, sdel_r, equal_nan=True)
+    assert np.allclose(umtx, umtx_r, equal_nan=True)
+
+
+def test_local_compute(small_case):
+    g = small_case
+    rng = np.random.RandomState(3)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, rng)
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    _, _, dproc = env.dmtx_baseline()
+    dec, dly = env.local_compute(dproc)
+    assert np.array_equal(dec, jobs.sources)
+    assert np.allclose(dly, np.maximum(dproc[jobs.sources] * jobs.ul, 1))
+    ldel, sdel, _ = env.run()
+    assert np.all(np.isnan(ldel))      # no links used
+    emp = delay_empirical(ldel, sdel)
+    assert emp.shape == (jobs.num_jobs,)
+    assert np.all(emp > 0)
+
+
+def test_softmax_matches_reference_quirk():
+    # util.py:113-116: plain exp-normalise, so HIGH costs get HIGH probability
+    x = np.array([0.0, 1.0, 2.0])
+    p = softmax(x)
+    assert p[2] > p[1] > p[0]
+    assert np.isclose(p.sum(), 1.0)
+
+
+def test_explore_and_prob_modes(small_case):
+    g = small_case
+    rng = np.random.RandomState(4)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, rng)
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    dmtx, dlist, dproc = env.dmtx_baseline()
+    sp = apsp(g, dlist)
+    np.fill_diagonal(sp, np.where(dproc > 0, dproc, g.T))
+    # explore=1 → uniformly random choices among servers+local
+    dec, _ = env.offloading(sp, g.sp_hop, explore=1.0,
+                            rng=np.random.RandomState(0))
+    assert all(d in set(g.servers) | {int(s)}
+               for d, s in zip(dec, jobs.sources))
+    # prob mode runs and picks valid destinations
+    dec, _ = env.offloading(sp, g.sp_hop, prob=True,
+                            rng=np.random.RandomState(0))
+    assert all(d in set(g.servers) | {int(s)}
+               for d, s in zip(dec, jobs.sources))
+
+
+def test_greedy_route_terminates_and_is_valid(small_case):
+    g = small_case
+    rng = np.random.RandomState(5)
+    w = rng.uniform(0.01, 1.0, g.num_links)
+    sp = apsp(g, w)
+    env = AdhocCloudEnv(g)
+    for src in g.mobile_nodes[:5]:
+        for dst in g.servers:
+            route, nh = env.routing(int(src), int(dst), sp)
+            assert route[0] == src and route[-1] == dst and nh == len(route) - 1
+            for a, b in zip(route[:-1], route[1:]):
+                assert g.adj[a, b] == 1

@@ -1,0 +1,128 @@
+"""Graph data layer vs networkx oracles."""
+import numpy as np
+import networkx as nx
+import pytest
+
+from multihop_offload_amd import CaseGraph, JobInstance
+
+
+def _nx_graph(g):
+    return nx.from_numpy_array(g.adj.astype(float))
+
+
+def test_conflict_graph_matches_line_graph(small_case):
+    g = small_case
+    G = _nx_graph(g)
+    L = nx.line_graph(G)
+    # map our canonical link ids to nx line-graph nodes
+    by_pair = {tuple(sorted(e)): i for i, e in enumerate(map(tuple, g.edges))}
+    assert len(L.nodes) == g.num_links
+    for (a, b) in L.nodes:
+        i = by_pair[tuple(sorted((a, b)))]
+        nbrs = {by_pair[tuple(sorted(x))] for x in L.neighbors((a, b))}
+        lo, hi = g.conf_indptr[i], g.conf_indptr[i + 1]
+        assert set(g.conf_indices[lo:hi].tolist()) == nbrs
+    # conflict degrees
+    degs = np.array([L.degree(n) for n in L.nodes])
+    assert np.sum(degs) == np.sum(g.cf_degs)
+
+
+def test_extended_graph_counts(small_case):
+    g = small_case
+    ext = g.ext
+    n_comp = int(np.sum(g.roles < 2))
+    assert ext.num_edges_ext == g.num_links + n_comp
+    # virtual-edge maps cover exactly the computing nodes, ascending
+    assert np.array_equal(ext.comp_nodes, np.nonzero(g.roles < 2)[0])
+    assert np.array_equal(ext.maps_on_el,
+                          g.num_links + np.arange(n_comp))
+    # a virtual edge conflicts with exactly the real links at its node
+    for k, u in enumerate(ext.comp_nodes):
+        e = g.num_links + k
+        lo, hi = ext.ext_indptr[e], ext.ext_indptr[e + 1]
+        got = set(ext.ext_indices[lo:hi].tolist())
+        want = set(g.link_matrix[u][g.adj[u] != 0].tolist())
+        assert got == want
+
+
+def test_extended_line_graph_matches_networkx(small_case):
+    """Full structural check of the extended line graph against the
+    reference construction (offloading_v3.py:262-339 semantics)."""
+    g = small_case
+    G = _nx_graph(g)
+    N = g.num_nodes
+    Ge = G.copy()
+    for idx in range(N):
+        if g.roles[idx] < 2:
+            Ge.add_edge(idx, N + idx)
+    L = nx.line_graph(Ge)
+    ext = g.ext
+
+    def canon_id(pair):
+        a, b = sorted(pair)
+        if b >= N:
+            return int(ext.node_vedge[a])
+        return int(g.link_matrix[a, b])
+
+    assert len(L.nodes) == ext.num_edges_ext
+    for node in L.nodes:
+        i = canon_id(node)
+        nbrs = {canon_id(x) for x in L.neighbors(node)}
+        lo, hi = ext.ext_indptr[i], ext.ext_indptr[i + 1]
+        assert set(ext.ext_indices[lo:hi].tolist()) == nbrs
+
+
+def test_features(small_case, jobs_for):
+    g, jobs = small_case, jobs_for
+    x = g.ext.features(jobs)
+    E = g.num_links
+    assert x.shape == (g.ext.num_edges_ext, 4)
+    assert np.all(x[:E, 0] == 0) and np.all(x[E:, 0] == 1)
+    assert np.allclose(x[:E, 1], g.link_rates)
+    assert np.allclose(x[E:, 1], g.proc_bws[g.ext.comp_nodes])
+    # job arrivals sit on the source nodes' virtual edges only
+    assert np.all(x[:E, 2] == 0)
+    per_node = np.zeros(g.num_nodes)
+    np.add.at(per_node, jobs.sources, jobs.rates * jobs.ul)
+    assert np.allclose(x[E:, 2], per_node[g.ext.comp_nodes])
+    # server flags
+    servers = set(g.servers)
+    for k, u in enumerate(g.ext.comp_nodes):
+        assert x[E + k, 3] == (1.0 if u in servers else 0.0)
+
+
+def test_job_sampling_distribution():
+    g = CaseGraph(30, seed=3, gtype="ba")
+    rng = np.random.RandomState(0)
+    mobiles = np.arange(5, 30)
+    for _ in range(20):
+        jobs = JobInstance.sample(mobiles, 0.15, rng)
+        assert int(0.3 * len(mobiles)) <= jobs.num_jobs < len(mobiles)
+        assert np.all(np.isin(jobs.sources, mobiles))
+        assert len(np.unique(jobs.sources)) == jobs.num_jobs
+        assert np.all((jobs.rates >= 0.1 * 0.15) & (jobs.rates <= 0.5 * 0.15))
+        assert np.all(jobs.ul == 100) and np.all(jobs.dl == 1)
+
+
+@pytest.mark.parametrize("gtype", ["ba", "er", "ws", "poisson"])
+def test_graph_families(gtype):
+    g = CaseGraph(30, seed=11, gtype=gtype, m=4 if gtype == "poisson" else 2)
+    assert g.num_links > 0
+    assert g.adj.sum() == 2 * g.num_links
+
+
+def test_mat_roundtrip(tmp_path):
+    from multihop_offload_amd.datagen import generate_case
+    import scipy.io as sio
+    rng = np.random.RandomState(5)
+    case, ns = generate_case(20, 123, "ba", rng=rng)
+    path = str(tmp_path / "case.mat")
+    sio.savemat(path, case)
+    g = CaseGraph.from_mat(path)
+    assert g.num_nodes == 20 and g.seed == 123 and g.m == 2
+    assert g.mat_link_rate.shape == (g.num_links,)
+    roles = case["nodes_info"][:, 0]
+    assert np.array_equal(g.roles, roles)
+    assert np.allclose(g.proc_bws, case["nodes_info"][:, 1])
+    assert len(g.servers) == int(np.sum(roles == 1))
+    assert len(g.relays) == int(np.sum(roles == 2))

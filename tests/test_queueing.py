@@ -1,0 +1,99 @@
+"""Queueing math vs literal NumPy transcriptions of the reference equations
+(offloading_v3.py:498-506, gnn_offloading_agent.py:240-254) + gradient checks."""
+import numpy as np
+import torch
+
+from multihop_offload_amd.queueing import (
+    ConflictCSR, actor_delays, delay_matrix, delay_with_fallback,
+    fixed_point_mu)
+
+
+def _naive_fixed_point(lam, rates, cf_degs, adj_dense, iters=10):
+    mu = rates / (cf_degs + 1.0)
+    for _ in range(iters):
+        busy = np.clip(lam / mu, 0, 1.0)
+        nb = adj_dense @ busy
+        mu = rates / (1.0 + nb)
+    return mu
+
+
+def _conf_dense(g):
+    E = g.num_links
+    a = np.zeros((E, E))
+    rows = np.repeat(np.arange(E), np.diff(g.conf_indptr))
+    a[rows, g.conf_indices] = 1.0
+    return a
+
+
+def test_fixed_point_matches_naive(small_case):
+    g = small_case
+    rng = np.random.RandomState(0)
+    lam = rng.uniform(0, 60, g.num_links)
+    conf = ConflictCSR(g.conf_indptr, g.conf_indices)
+    mu = fixed_point_mu(torch.tensor(lam), torch.tensor(g.link_rates),
+                        torch.tensor(g.cf_degs), conf)
+    want = _naive_fixed_point(lam, g.link_rates, g.cf_degs, _conf_dense(g))
+    assert np.allclose(mu.numpy(), want, rtol=1e-12)
+
+
+def test_delay_fallback_branches():
+    lam = torch.tensor([1.0, 5.0, 10.0], dtype=torch.float64)
+    mu = torch.tensor([4.0, 5.0, 5.0], dtype=torch.float64)
+    d = delay_with_fallback(lam, mu, T=1000, denom=101.0)
+    assert np.isclose(d[0].item(), 1.0 / 3.0)
+    # tie lam == mu is NOT congested (strict >) → 1/0 = inf like the reference
+    assert np.isinf(d[1].item())
+    assert np.isclose(d[2].item(), 1000.0 * 10.0 / (101.0 * 5.0))
+
+
+def test_fixed_point_gradcheck(small_case):
+    g = small_case
+    conf = ConflictCSR(g.conf_indptr, g.conf_indices)
+    rates = torch.tensor(g.link_rates)
+    cf = torch.tensor(g.cf_degs)
+    lam = torch.tensor(np.random.RandomState(1).uniform(1, 30, g.num_links),
+                       requires_grad=True)
+
+    def f(x):
+        return fixed_point_mu(x, rates, cf, conf).sum()
+
+    assert torch.autograd.gradcheck(f, (lam,), eps=1e-6, atol=1e-6)
+
+
+def test_delay_matrix_scatter(small_case):
+    g = small_case
+    ext = g.ext
+    E = g.num_links
+    rng = np.random.RandomState(2)
+    ld = torch.tensor(rng.uniform(0.1, 2, E))
+    nd = torch.tensor(rng.uniform(0.1, 2, len(ext.comp_nodes)))
+    dm = delay_matrix(ld, nd, torch.tensor(g.edges),
+                      torch.tensor(ext.comp_nodes), g.num_nodes).numpy()
+    for l, (u, v) in enumerate(g.edges):
+        assert dm[u, v] == dm[v, u] == ld[l].item()
+    for k, u in enumerate(ext.comp_nodes):
+        assert dm[u, u] == nd[k].item()
+    for r in np.nonzero(g.roles == 2)[0]:
+        assert np.isinf(dm[r, r])
+    # non-edges are zero
+    off = dm.copy()
+    off[g.edges[:, 0], g.edges[:, 1]] = 0
+    off[g.edges[:, 1], g.edges[:, 0]] = 0
+    np.fill_diagonal(off, 0)
+    assert np.all(off == 0)
+
+
+def test_actor_delays_pipeline_gradflow(small_case):
+    g = small_case
+    ext = g.ext
+    conf = ConflictCSR(g.conf_indptr, g.conf_indices)
+    E, C = g.num_links, len(ext.comp_nodes)
+    rng = np.random.RandomState(3)
+    lam_l = torch.tensor(rng.uniform(0, 20, E), requires_grad=True)
+    lam_n = torch.tensor(rng.uniform(0, 5, C), requires_grad=True)
+    ld, nd = actor_delays(lam_l, lam_n, torch.tensor(g.link_rates),
+                          torch.tensor(g.cf_degs), conf,
+                          torch.tensor(g.proc_bws[ext.comp_nodes]), 1000)
+    (ld.sum() + nd.sum()).backward()
+    assert torch.isfinite(lam_l.grad).all()
+    assert torch.isfinite(lam_n.grad).all()
