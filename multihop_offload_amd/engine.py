@@ -1,0 +1,613 @@
+"""Batched device-resident episode engine.
+
+The reference processes one graph × one job instance at a time, with Python
+loops over jobs/links and CPU Dijkstra between two TF segments
+(SURVEY.md §3.1).  This engine runs **B (graph, jobs) instances as one tensor
+program** — features → ChebConv → contention fixed point → delay matrices →
+batched min-plus APSP → decision/greedy-routing walk → analytic evaluation →
+critic → semi-analytic backward — with zero host round-trips inside a step.
+On MI355X the hot stages dispatch to HIP kernels (ops/); the same code runs
+on CPU (fp64) where it is tested for exact agreement with the oracle
+``AdhocCloudEnv``/``ACOAgent`` path.
+
+Batching layout:
+  * per-graph dense tensors are stacked: (B, N, N) / (B, E) / (B, Ē);
+    all graphs in a batch share N (and hence E for BA(m)); ragged
+    conflict structure is flat block-diagonal CSR over B·E (resp. B·Ē);
+  * job instances are padded to Jmax with a validity mask; padded jobs have
+    zero rate and are excluded from routes, losses and metrics;
+  * the per-instance actor gradient sets of the reference's replay memory
+    (gnn_offloading_agent.py:156-169) become one fused VJP: instances are
+    independent blocks, so the batched cotangent yields the SUM of the
+    per-instance gradients in a single backward pass.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from .graphs import CaseGraph
+from .models.chebconv import ChebConvStack
+from .queueing import ConflictCSR, delay_with_fallback, fixed_point_mu
+from .ops import dispatch as ops
+
+
+@dataclasses.dataclass
+class JobBatch:
+    sources: torch.Tensor   # (B, J) int64, padded with a valid mobile node
+    mask: torch.Tensor      # (B, J) bool — real jobs
+    rates: torch.Tensor     # (B, J)
+    ul: torch.Tensor        # (B, J)
+    dl: torch.Tensor        # (B, J)
+
+    @property
+    def num_real(self):
+        return self.mask.sum()
+
+
+@dataclasses.dataclass
+class EpisodeResult:
+    tau: torch.Tensor            # (B,) mean per-job empirical delay
+    congest: torch.Tensor        # (B,) number of congested jobs
+    num_jobs: torch.Tensor       # (B,)
+    delay_emp: torch.Tensor      # (B, J) per-job empirical delay (nan-padded)
+    loss_fn: Optional[float] = None
+    loss_mse: Optional[float] = None
+
+
+class EpisodeEngine:
+    def __init__(self, cases: Sequence[CaseGraph], model: ChebConvStack,
+                 device: str = "cpu", dtype: torch.dtype = torch.float32,
+                 fp_iters: int = 10, walk_cap: Optional[int] = None):
+        assert len(cases) > 0
+        N = cases[0].num_nodes
+        E = cases[0].num_links
+        T = cases[0].T
+        assert all(c.num_nodes == N for c in cases), "batch shares N"
+        assert all(c.num_links == E for c in cases), "batch shares E"
+        self.cases = list(cases)
+        self.B, self.N, self.E, self.T = len(cases), N, E, T
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.fp_iters = fp_iters
+        self.model = model.to(self.device)
+        self.walk_cap = walk_cap or N
+
+        # extended edge dimension is ragged across graphs (relay counts
+        # differ) — pad to the max; padded edges are isolated in the support
+        # and excluded from every scatter by the comp-index tables below
+        Ee = max(c.ext.num_edges_ext for c in cases)
+        self.Ee = Ee
+        B = self.B
+
+        def t(a, dt=None):
+            return torch.as_tensor(np.asarray(a), dtype=dt or dtype,
+                                   device=self.device)
+
+        def ti(a):
+            return torch.as_tensor(np.asarray(a), dtype=torch.int64,
+                                   device=self.device)
+
+        self.edges = ti([c.edges for c in cases])              # (B,E,2)
+        self.link_rates = t([c.link_rates for c in cases])     # (B,E)
+        self.cf_degs = t([c.cf_degs for c in cases])
+        self.proc_bws = t([c.proc_bws for c in cases])         # (B,N)
+        self.roles = ti([c.roles for c in cases])
+        self.adj = torch.as_tensor(
+            np.stack([c.adj for c in cases]) != 0, device=self.device)
+        self.link_matrix = ti([c.link_matrix for c in cases])  # (B,N,N)
+        self.node_vedge = ti([c.ext.node_vedge for c in cases])  # (B,N), -1 relay
+        self.comp_mask = self.roles < 2                        # (B,N)
+        self.sp_hop = t([c.sp_hop for c in cases])             # (B,N,N)
+
+        # servers, padded
+        S = max(len(c.servers) for c in cases)
+        srv = -np.ones((B, S), dtype=np.int64)
+        for b, c in enumerate(cases):
+            srv[b, :len(c.servers)] = c.servers
+        self.S = S
+        self.servers = ti(srv)
+        self.server_mask = self.servers >= 0
+        self.servers_safe = self.servers.clamp(min=0)
+
+        # mobiles
+        mob_counts = [len(c.mobile_nodes) for c in cases]
+        self.Jmax = max(mob_counts) - 1 if max(mob_counts) > 1 else 1
+        self.mobile_mask = self.roles == 0
+
+        # flat block-diagonal conflict CSR over B·E (original line graph)
+        rows, cols = [], []
+        for b, c in enumerate(cases):
+            r = np.repeat(np.arange(E), np.diff(c.conf_indptr)) + b * E
+            rows.append(r)
+            cols.append(c.conf_indices + b * E)
+        self.conf = ConflictCSR.__new__(ConflictCSR)
+        self.conf.row = ti(np.concatenate(rows))
+        self.conf.col = ti(np.concatenate(cols))
+        self.conf.n = B * E
+
+        # flat block-diagonal extended line-graph CSR over B·Ē (GNN support)
+        rows, cols = [], []
+        for b, c in enumerate(cases):
+            ext = c.ext
+            ne = ext.num_edges_ext            # ≤ Ee; padded tail is isolated
+            r = np.repeat(np.arange(ne), np.diff(ext.ext_indptr)) + b * Ee
+            rows.append(r)
+            cols.append(ext.ext_indices + b * Ee)
+        self.support = ConflictCSR.__new__(ConflictCSR)
+        self.support.row = ti(np.concatenate(rows))
+        self.support.col = ti(np.concatenate(cols))
+        self.support.n = B * Ee
+
+        # static feature columns (B,Ē), padded (self_loop=1, rate=1)
+        def pad(vals, fill):
+            out = np.full((B, Ee), fill, dtype=np.float64)
+            for b, v in enumerate(vals):
+                out[b, :len(v)] = v
+            return t(out)
+
+        self.f_self_loop = pad([c.ext.edge_self_loop for c in cases], 1.0)
+        self.f_as_server = pad([c.ext.edge_as_server for c in cases], 0.0)
+        for c in cases:
+            c.ext.refresh_rates()
+        self.f_rate = pad([c.ext.edge_rate_ext for c in cases], 1.0)
+        # proc bws of computing nodes in extended-edge order (B, Ē-E),
+        # padded with 1.0 (kept finite; padded slots never reach a scatter)
+        self.Cmax = Ee - E
+        bw = np.ones((B, self.Cmax), dtype=np.float64)
+        for b, c in enumerate(cases):
+            bw[b, :len(c.ext.comp_nodes)] = c.proc_bws[c.ext.comp_nodes]
+        self.bw_comp = t(bw)
+
+        # computing-node index tables: for every (graph b, computing node n):
+        # its node id, its rank in the graph's extended-edge tail
+        cb, cn, cr = [], [], []
+        for b, c in enumerate(cases):
+            for rank, n in enumerate(c.ext.comp_nodes):
+                cb.append(b)
+                cn.append(int(n))
+                cr.append(rank)
+        self._comp_b = ti(cb)
+        self._comp_n = ti(cn)
+        self._comp_r = ti(cr)
+
+        self._bidx = torch.arange(B, device=self.device)
+
+    # ------------------------------------------------------------------ jobs
+    def sample_jobs(self, arrival_scale: float,
+                    gen: Optional[torch.Generator] = None) -> JobBatch:
+        """Device-side analog of ``JobInstance.sample``: per graph, J ~
+        U{int(0.3·M), M-1} jobs on a random subset of mobiles, rates
+        U(0.1,0.5)·scale, ul=100, dl=1."""
+        B, N, J = self.B, self.N, self.Jmax
+        dev = self.device
+        scores = torch.rand(B, N, device=dev, generator=gen)
+        scores = torch.where(self.mobile_mask, scores,
+                             torch.full_like(scores, 2.0))
+        order = torch.argsort(scores, dim=1)                 # mobiles first
+        sources = order[:, :J]                               # (B,J)
+        M = self.mobile_mask.sum(1)                          # (B,)
+        lo = (0.3 * M).to(torch.int64)
+        span = (M - lo).clamp(min=1)
+        nj = lo + (torch.rand(B, device=dev, generator=gen)
+                   * span.to(self.dtype)).to(torch.int64)
+        nj = nj.clamp(min=1)
+        mask = torch.arange(J, device=dev)[None, :] < nj[:, None]
+        # clamp padded sources to a valid mobile (first sampled) for safety
+        sources = torch.where(mask, sources, sources[:, :1])
+        rates = (torch.rand(B, J, device=dev, generator=gen) * 0.4 + 0.1) \
+            * arrival_scale
+        z = torch.zeros(B, J, device=dev, dtype=self.dtype)
+        return JobBatch(
+            sources=sources,
+            mask=mask,
+            rates=torch.where(mask, rates.to(self.dtype), z),
+            ul=torch.where(mask, torch.full_like(z, 100.0), z),
+            dl=torch.where(mask, torch.full_like(z, 1.0), z),
+        )
+
+    # ------------------------------------------------------- actor forward
+    def actor_forward(self, jobs: JobBatch):
+        """features → ChebConv → λ → fixed point → delays → (B,N,N) delay
+        matrix (autograd graph kept).  Reference gnn_offloading_agent.py:
+        211-276, batched."""
+        B, E, Ee, N = self.B, self.E, self.Ee, self.N
+        arr = torch.zeros(B, N, dtype=self.dtype, device=self.device)
+        arr = arr.scatter_add(1, jobs.sources, jobs.rates * jobs.ul)
+        f_job = torch.zeros(B, Ee, dtype=self.dtype, device=self.device)
+        vidx = torch.where(self.comp_mask, self.node_vedge,
+                           torch.zeros_like(self.node_vedge))
+        f_job = f_job.scatter_add(
+            1, vidx, torch.where(self.comp_mask, arr, torch.zeros_like(arr)))
+
+        x = torch.stack([self.f_self_loop, self.f_rate, f_job,
+                         self.f_as_server], dim=-1).reshape(B * Ee, 4)
+        lam = self.model(x, self.support).reshape(B, Ee)
+
+        lam_link = lam[:, :E].reshape(-1)
+        mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),
+                            self.cf_degs.reshape(-1), self.conf,
+                            self.fp_iters)
+        link_delay = delay_with_fallback(lam_link, mu, self.T, 101.0) \
+            .reshape(B, E)
+        node_delay = delay_with_fallback(lam[:, E:], self.bw_comp,
+                                         self.T, 100.0)       # (B, Ē-E)
+
+        dm = self._delay_matrix(link_delay, node_delay)
+        return dm, link_delay, node_delay
+
+    def _delay_matrix(self, link_delay, node_delay):
+        B, N, E = self.B, self.N, self.E
+        flat = torch.zeros(B * N * N, dtype=self.dtype, device=self.device)
+        e0, e1 = self.edges[..., 0], self.edges[..., 1]
+        base = self._bidx[:, None] * (N * N)
+        flat = flat.index_put(((base + e0 * N + e1).reshape(-1),),
+                              link_delay.reshape(-1))
+        flat = flat.index_put(((base + e1 * N + e0).reshape(-1),),
+                              link_delay.reshape(-1))
+        # diagonal: node delays at computing nodes, +inf at relays
+        diag_idx = base + torch.arange(N, device=self.device)[None, :] * (N + 1)
+        inf = torch.full((B, N), float("inf"), dtype=self.dtype,
+                         device=self.device)
+        flat = flat.index_put((diag_idx.reshape(-1),), inf.reshape(-1))
+        lin = (self._comp_b * (N * N) + self._comp_n * (N + 1))
+        flat = flat.index_put((lin,), node_delay[self._comp_b, self._comp_r])
+        return flat.reshape(B, N, N)
+
+    # --------------------------------------------------------------- APSP
+    def apsp(self, dm: torch.Tensor) -> torch.Tensor:
+        """Batched all-pairs shortest paths over the delay-weighted graphs.
+        Input: (B,N,N) delay matrix (off-diagonal link delays; diagonal
+        ignored).  Non-edges → +inf, diagonal → 0, then min-plus
+        Floyd–Warshall (detached — the reference's Dijkstra is outside the
+        tape too, gnn_offloading_agent.py:304-306)."""
+        B, N = self.B, self.N
+        with torch.no_grad():
+            w = torch.where(self.adj, dm.detach(),
+                            torch.full_like(dm, float("inf")))
+            idx = torch.arange(N, device=self.device)
+            w[:, idx, idx] = 0.0
+            return ops.floyd_warshall(w)
+
+    # ----------------------------------------------- decision + routing walk
+    def offload_decide(self, jobs: JobBatch, sp: torch.Tensor,
+                       uds: torch.Tensor, explore: float = 0.0,
+                       gen: Optional[torch.Generator] = None,
+                       prob: bool = False):
+        """Greedy server selection (offloading_v3.py:388-439), batched.
+        ``sp``: (B,N,N) shortest-path delays with ZERO diagonal;
+        ``uds``: (B,N) per-node unit processing delays (inf at relays)."""
+        B, J, S = self.B, self.Jmax, self.S
+        bJ = self._bidx[:, None]
+        src = jobs.sources                                    # (B,J)
+        local = uds.gather(1, src) * jobs.ul                  # (B,J)
+
+        sp_src = sp[bJ, src]                                  # (B,J,N)
+        hop_src = self.sp_hop[bJ, src]                        # (B,J,N)
+        srv = self.servers_safe[:, None, :].expand(B, J, S)
+        ul_d = torch.maximum(sp_src.gather(2, srv) * jobs.ul[..., None],
+                             hop_src.gather(2, srv))
+        # undirected graphs: sp/hop are symmetric, so dl leg reuses the rows
+        dl_d = torch.maximum(sp_src.gather(2, srv) * jobs.dl[..., None],
+                             hop_src.gather(2, srv))
+        uds_srv = uds.gather(1, srv.reshape(B, -1)).reshape(B, J, S)
+        pr_d = torch.maximum(uds_srv * jobs.ul[..., None],
+                             torch.ones_like(uds_srv))
+        server_costs = ul_d + dl_d + pr_d                     # (B,J,S)
+        server_costs = torch.where(
+            self.server_mask[:, None, :], server_costs,
+            torch.full_like(server_costs, float("inf")))
+        costs = torch.cat([server_costs, local[..., None]], dim=2)  # (B,J,S+1)
+
+        if prob:
+            logits = costs.clamp(max=1e30)
+            logits = logits - logits.amax(dim=2, keepdim=True)
+            p = torch.exp(logits)
+            p = torch.where(torch.isfinite(costs), p, torch.zeros_like(p))
+            choice = torch.multinomial(
+                p.reshape(B * J, S + 1), 1, generator=gen).reshape(B, J)
+        else:
+            choice = costs.argmin(dim=2)                      # (B,J)
+        if explore > 0:
+            nS = self.server_mask.sum(1, keepdim=True)        # (B,1)
+            r = torch.rand(B, J, device=self.device, generator=gen)
+            rc = (torch.rand(B, J, device=self.device, generator=gen)
+                  * (nS + 1).to(self.dtype)).to(torch.int64)
+            # uniform over [0..S_b]; index S_b means "local" → map to slot S
+            rc = torch.where(rc >= nS, torch.full_like(rc, S), rc)
+            choice = torch.where(r < explore, rc, choice)
+
+        is_local = choice >= torch.minimum(
+            self.server_mask.sum(1, keepdim=True),
+            torch.full_like(choice, S))
+        # map server slots through the server table
+        dst = torch.where(
+            is_local, src,
+            self.servers_safe.gather(1, choice.clamp(max=S - 1)))
+        dst = torch.where(jobs.mask, dst, src)
+        est = torch.where(
+            is_local, local,
+            server_costs.gather(2, choice.clamp(max=S - 1)[..., None])[..., 0])
+        return dst, est
+
+    def route_walk(self, jobs: JobBatch, dst: torch.Tensor, sp: torch.Tensor):
+        """Greedy next-hop walk (offloading_v3.py:441-453), lockstep over all
+        jobs: next = argmin over neighbors of sp[nb, dst] (first-min =
+        ascending node id, like np.argmin).  Returns (route_links (B,J,H)
+        int64 padded -1, nhop (B,J))."""
+        B, J, N = self.B, self.Jmax, self.N
+        bJ = self._bidx[:, None]
+        node = jobs.sources.clone()
+        links, hops = [], torch.zeros(B, J, dtype=torch.int64,
+                                      device=self.device)
+        sp_to_dst = sp.gather(
+            2, dst[:, None, :].expand(B, N, J))               # (B,N,J)
+        for _ in range(self.walk_cap):
+            active = (node != dst) & jobs.mask
+            if not bool(active.any()):
+                break
+            adj_row = self.adj[bJ, node]                      # (B,J,N)
+            cand = torch.where(adj_row, sp_to_dst.transpose(1, 2),
+                               torch.full_like(sp_to_dst.transpose(1, 2),
+                                               float("inf")))
+            nxt = cand.argmin(dim=2)                          # (B,J)
+            link = self.link_matrix[bJ, node, nxt]            # (B,J)
+            links.append(torch.where(active, link,
+                                     torch.full_like(link, -1)))
+            node = torch.where(active, nxt, node)
+            hops += active.to(torch.int64)
+        if links:
+            route_links = torch.stack(links, dim=2)           # (B,J,H)
+        else:
+            route_links = torch.full((B, J, 0), -1, dtype=torch.int64,
+                                     device=self.device)
+        return route_links, hops
+
+    # --------------------------------------------------------- evaluation
+    def evaluate(self, jobs: JobBatch, dst: torch.Tensor,
+                 route_links: torch.Tensor, nhop: torch.Tensor):
+        """Analytic queueing evaluation (offloading_v3.py:455-550), batched.
+        Returns (delay_emp (B,J), unit_mtx (B,N,N), unit_mask, link_lambda,
+        link_mu, server_load)."""
+        B, E, N, J = self.B, self.E, self.N, self.Jmax
+        H = route_links.shape[2]
+        valid = route_links >= 0                              # (B,J,H)
+        safe = route_links.clamp(min=0)
+
+        ulr = jobs.ul * jobs.rates
+        dlr = jobs.dl * jobs.rates
+        load = (ulr + dlr)[..., None].expand(B, J, H)
+        flat_idx = (self._bidx[:, None, None] * E + safe).reshape(-1)
+        link_lambda = torch.zeros(B * E, dtype=self.dtype, device=self.device)
+        link_lambda = link_lambda.scatter_add(
+            0, flat_idx, torch.where(valid, load,
+                                     torch.zeros_like(load)).reshape(-1))
+        server_load = torch.zeros(B, N, dtype=self.dtype, device=self.device)
+        server_load = server_load.scatter_add(
+            1, dst, torch.where(jobs.mask, ulr, torch.zeros_like(ulr)))
+
+        with torch.no_grad():
+            mu = fixed_point_mu(link_lambda, self.link_rates.reshape(-1),
+                                self.cf_degs.reshape(-1), self.conf,
+                                self.fp_iters)
+        lam_r = link_lambda.reshape(B, E)
+        mu_r = mu.reshape(B, E)
+
+        # per-(job,hop) unit link delays with the per-job congestion fallback
+        lam_h = lam_r.gather(1, safe.reshape(B, -1)).reshape(B, J, H)
+        mu_h = mu_r.gather(1, safe.reshape(B, -1)).reshape(B, J, H)
+        tot = (jobs.ul + jobs.dl)[..., None]
+        gap = mu_h - lam_h
+        unit = torch.where(gap > 0, 1.0 / torch.where(gap > 0, gap,
+                                                      torch.ones_like(gap)),
+                           float(self.T) * lam_h / (tot * mu_h))
+        nh = nhop.to(self.dtype)[..., None]
+        hop_delay = (torch.maximum(jobs.ul[..., None] * unit, nh)
+                     + torch.maximum(jobs.dl[..., None] * unit, nh))
+        link_part = torch.where(valid, hop_delay,
+                                torch.zeros_like(hop_delay)).sum(dim=2)
+
+        bw_dst = self.proc_bws.gather(1, dst)
+        sl_dst = server_load.gather(1, dst)
+        sgap = bw_dst - sl_dst
+        sunit = torch.where(
+            sgap > 0, 1.0 / torch.where(sgap > 0, sgap, torch.ones_like(sgap)),
+            float(self.T) * sl_dst / (jobs.ul * bw_dst))
+        server_part = torch.maximum(jobs.ul * sunit, torch.ones_like(sunit))
+        delay_emp = torch.where(jobs.mask, link_part + server_part,
+                                torch.full_like(link_part, float("nan")))
+
+        # empirical unit-delay matrix + mask (for the MSE anchor)
+        unit_mtx = torch.zeros(B * N * N, dtype=self.dtype, device=self.device)
+        written = torch.zeros(B * N * N, dtype=torch.bool, device=self.device)
+        e0 = self.edges[..., 0].gather(1, safe.reshape(B, -1)).reshape(B, J, H)
+        e1 = self.edges[..., 1].gather(1, safe.reshape(B, -1)).reshape(B, J, H)
+        base = (self._bidx[:, None, None] * (N * N))
+        lin01 = (base + e0 * N + e1).reshape(-1)
+        lin10 = (base + e1 * N + e0).reshape(-1)
+        vmask = valid.reshape(-1)
+        unit_mtx = unit_mtx.index_put((lin01[vmask],),
+                                      unit.reshape(-1)[vmask])
+        unit_mtx = unit_mtx.index_put((lin10[vmask],),
+                                      unit.reshape(-1)[vmask])
+        written = written.index_put((lin01[vmask],),
+                                    torch.ones_like(lin01[vmask],
+                                                    dtype=torch.bool))
+        written = written.index_put((lin10[vmask],),
+                                    torch.ones_like(lin10[vmask],
+                                                    dtype=torch.bool))
+        lin_d = (self._bidx[:, None] * (N * N) + dst * (N + 1)).reshape(-1)
+        jm = jobs.mask.reshape(-1)
+        unit_mtx = unit_mtx.index_put((lin_d[jm],), sunit.reshape(-1)[jm])
+        written = written.index_put((lin_d[jm],),
+                                    torch.ones_like(lin_d[jm],
+                                                    dtype=torch.bool))
+        return (delay_emp, unit_mtx.reshape(B, N, N),
+                written.reshape(B, N, N), lam_r, mu_r, server_load)
+
+    # ------------------------------------------------------------- critic
+    def critic_backward(self, jobs: JobBatch, dst: torch.Tensor,
+                        route_links: torch.Tensor):
+        """Critic loss over the routes tensor + grad wrt routes
+        (gnn_offloading_agent.py:333-374) and the closed-form route-bias
+        cotangent accumulation (:384-416), batched.  Returns
+        (grad_edge (B,Ē), loss_fn scalar)."""
+        B, E, Ee, J = self.B, self.E, self.Ee, self.Jmax
+        H = route_links.shape[2]
+        valid = route_links >= 0
+        safe = route_links.clamp(min=0)
+        vedge_dst = self.node_vedge.gather(1, dst)            # (B,J)
+
+        # routes (B,Ē,J) — link hops + destination self-loop edge
+        routes = torch.zeros(B, Ee, J, dtype=self.dtype, device=self.device)
+        jidx = torch.arange(J, device=self.device)[None, :, None] \
+            .expand(B, J, H)
+        routes = routes.index_put(
+            (self._bidx[:, None, None].expand(B, J, H)[valid],
+             safe[valid], jidx[valid]), torch.ones(1, dtype=self.dtype,
+                                                   device=self.device)[0]
+            .expand(int(valid.sum())))
+        jm = jobs.mask
+        routes = routes.index_put(
+            (self._bidx[:, None].expand(B, J)[jm], vedge_dst[jm],
+             torch.arange(J, device=self.device)[None, :].expand(B, J)[jm]),
+            torch.ones(int(jm.sum()), dtype=self.dtype, device=self.device))
+        routes.requires_grad_(True)
+
+        jobs_load = (jobs.rates * jobs.ul)[..., None]          # (B,J,1)
+        jobs_data = (jobs.ul + jobs.dl)[:, None, :]            # (B,1,J)
+        link_load = torch.bmm(routes, jobs_load)[..., 0]       # (B,Ē)
+        lam_link = link_load[:, :E].reshape(-1)
+        mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),
+                            self.cf_degs.reshape(-1), self.conf,
+                            self.fp_iters)
+        link_d = delay_with_fallback(lam_link, mu, self.T, 101.0) \
+            .reshape(B, E)
+        node_d = delay_with_fallback(link_load[:, E:], self.bw_comp,
+                                     self.T, 100.0)
+        unit_edge = torch.cat([link_d, node_d], dim=1)         # (B,Ē)
+        delay_job_edge = torch.maximum(
+            jobs_data * unit_edge[..., None] * routes, routes)
+        loss_fn = delay_job_edge.sum()
+        (grad_routes,) = torch.autograd.grad(loss_fn, routes)  # (B,Ē,J)
+
+        # route-bias VJP: forward prefix sums of -grad_routes along each route
+        # (links in order, then the destination self-loop edge)
+        seq = torch.cat([safe, vedge_dst[..., None]], dim=2)   # (B,J,H+1)
+        seq_valid = torch.cat([valid, jm[..., None]], dim=2)
+        g_seq = grad_routes.gather(
+            1, seq.transpose(1, 2)).transpose(1, 2)            # (B,J,H+1)? no:
+        # grad_routes is (B,Ē,J): gather along Ē with index (B,H+1,J) where
+        # index[b,h,j] = seq[b,j,h]
+        g_seq = grad_routes.gather(1, seq.permute(0, 2, 1))    # (B,H+1,J)
+        g_seq = g_seq.permute(0, 2, 1)                         # (B,J,H+1)
+        g_seq = torch.where(seq_valid, g_seq, torch.zeros_like(g_seq))
+        pref = -torch.cumsum(g_seq, dim=2)
+        grad_edge = torch.zeros(B * Ee, dtype=self.dtype, device=self.device)
+        flat_seq = (self._bidx[:, None, None] * Ee + seq).reshape(-1)
+        grad_edge = grad_edge.scatter_add(
+            0, flat_seq, torch.where(seq_valid, pref,
+                                     torch.zeros_like(pref)).reshape(-1))
+        return grad_edge.reshape(B, Ee), float(loss_fn.detach())
+
+    def grad_dist_matrix(self, grad_edge: torch.Tensor, dm: torch.Tensor,
+                         unit_mtx: torch.Tensor, written: torch.Tensor):
+        """Assemble the N×N actor cotangent: route-bias scatter
+        (gnn_offloading_agent.py:410-416) + 0.001·MSE anchor (:440-444)."""
+        B, N, E = self.B, self.N, self.E
+        flat = torch.zeros(B * N * N, dtype=self.dtype, device=self.device)
+        e0, e1 = self.edges[..., 0], self.edges[..., 1]
+        base = self._bidx[:, None] * (N * N)
+        flat = flat.index_put(((base + e0 * N + e1).reshape(-1),),
+                              grad_edge[:, :E].reshape(-1))
+        flat = flat.index_put(((base + e1 * N + e0).reshape(-1),),
+                              grad_edge[:, :E].reshape(-1))
+        lin = (self._comp_b * (N * N) + self._comp_n * (N + 1))
+        flat = flat.index_put(
+            (lin,), grad_edge[self._comp_b, E + self._comp_r])
+        grad_dist = flat.reshape(B, N, N)
+
+        diff = dm.detach() - unit_mtx
+        anchor_mask = written & torch.isfinite(dm.detach())
+        anchor = torch.where(anchor_mask, 0.001 * diff,
+                             torch.zeros_like(diff))
+        n_valid = int(anchor_mask.sum())
+        loss_mse = float((torch.where(anchor_mask, diff, torch.zeros_like(diff))
+                          ** 2).sum() / max(n_valid, 1))
+        return grad_dist + anchor, loss_mse
+
+    # ------------------------------------------------------------ episodes
+    def gnn_episode(self, jobs: JobBatch, explore: float = 0.0,
+                    gen: Optional[torch.Generator] = None,
+                    train: bool = True, prob: bool = False) -> EpisodeResult:
+        """One full GNN episode over the batch.  With ``train=True`` the
+        summed per-instance actor gradients are left in ``model.param.grad``
+        (caller applies the optimizer / DP all-reduce)."""
+        ctx = torch.enable_grad() if train else torch.no_grad()
+        with ctx:
+            dm, link_delay, node_delay = self.actor_forward(jobs)
+        sp = self.apsp(dm)
+        idx = torch.arange(self.N, device=self.device)
+        uds = dm.detach()[:, idx, idx]                        # (B,N)
+        dst, est = self.offload_decide(jobs, sp, uds, explore, gen, prob)
+        route_links, nhop = self.route_walk(jobs, dst, sp)
+        delay_emp, unit_mtx, written, lam_r, mu_r, server_load = \
+            self.evaluate(jobs, dst, route_links, nhop)
+
+        loss_fn = loss_mse = None
+        if train:
+            grad_edge, loss_fn = self.critic_backward(jobs, dst, route_links)
+            grad_dist, loss_mse = self.grad_dist_matrix(
+                grad_edge, dm, unit_mtx, written)
+            dm.backward(grad_dist)
+
+        nj = jobs.mask.sum(1)
+        de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
+        tau = de.sum(1) / nj.to(self.dtype)
+        congest = ((delay_emp > float(self.T)) & jobs.mask).sum(1)
+        return EpisodeResult(tau=tau, congest=congest, num_jobs=nj,
+                             delay_emp=delay_emp, loss_fn=loss_fn,
+                             loss_mse=loss_mse)
+
+    def baseline_episode(self, jobs: JobBatch) -> EpisodeResult:
+        """Greedy baseline (AdHoc_train.py:126-142), batched."""
+        B, N = self.B, self.N
+        with torch.no_grad():
+            # 1/rate stays +inf at rate==0, exactly like the reference's
+            # dead `delay if delay > 0 else T` guard (AdHoc_train.py:132)
+            dlist = 1.0 / self.link_rates                      # (B,E)
+            dm = self._delay_matrix(
+                dlist, torch.zeros_like(self.bw_comp))
+            sp = self.apsp(dm)
+            dproc = 1.0 / self.proc_bws
+            uds = torch.where((self.proc_bws > 0),
+                              dproc, torch.full_like(dproc, float("inf")))
+            dst, est = self.offload_decide(jobs, sp, uds)
+            route_links, nhop = self.route_walk(jobs, dst, sp)
+            delay_emp, *_ = self.evaluate(jobs, dst, route_links, nhop)
+            nj = jobs.mask.sum(1)
+            de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
+            return EpisodeResult(
+                tau=de.sum(1) / nj.to(self.dtype),
+                congest=((delay_emp > float(self.T)) & jobs.mask).sum(1),
+                num_jobs=nj, delay_emp=delay_emp)
+
+    def local_episode(self, jobs: JobBatch) -> EpisodeResult:
+        """Local computing (offloading_v3.py:363-386), batched."""
+        B, J = self.B, self.Jmax
+        with torch.no_grad():
+            dst = jobs.sources
+            route_links = torch.full((B, J, 0), -1, dtype=torch.int64,
+                                     device=self.device)
+            nhop = torch.zeros(B, J, dtype=torch.int64, device=self.device)
+            delay_emp, *_ = self.evaluate(jobs, dst, route_links, nhop)
+            nj = jobs.mask.sum(1)
+            de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
+            return EpisodeResult(
+                tau=de.sum(1) / nj.to(self.dtype),
+                congest=((delay_emp > float(self.T)) & jobs.mask).sum(1),
+                num_jobs=nj, delay_emp=delay_emp)

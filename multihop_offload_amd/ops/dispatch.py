@@ -1,0 +1,52 @@
+"""Device dispatch for the compute kernels.
+
+CPU tensors → pure-torch reference implementations (``torch_ref``).
+GPU tensors → the gfx950 HIP extension (``ops/hip``), which is REQUIRED on
+GPU: a missing extension raises instead of silently falling back to eager
+torch, so GPU runs always exercise the native kernels.
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+
+from . import torch_ref
+
+_HIP_EXT = None
+_HIP_TRIED = False
+
+
+def _load_hip():
+    global _HIP_EXT, _HIP_TRIED
+    if _HIP_TRIED:
+        return _HIP_EXT
+    _HIP_TRIED = True
+    try:
+        from . import hip_loader
+        _HIP_EXT = hip_loader.load()
+    except Exception:
+        _HIP_EXT = None
+    return _HIP_EXT
+
+
+def hip_available() -> bool:
+    return _load_hip() is not None
+
+
+def _require_hip():
+    ext = _load_hip()
+    if ext is None:
+        raise RuntimeError(
+            "multihop_offload_amd HIP extension not built — run "
+            "`python setup.py build_ext --inplace` (gfx950). GPU execution "
+            "without the native kernels is disabled by design.")
+    return ext
+
+
+def floyd_warshall(w: torch.Tensor) -> torch.Tensor:
+    if w.is_cuda and os.environ.get("MHO_FORCE_TORCH") != "1":
+        ext = _require_hip()
+        return ext.floyd_warshall(w.contiguous())
+    return torch_ref.floyd_warshall(w)

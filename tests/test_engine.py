@@ -1,0 +1,188 @@
+"""Batched engine vs the single-case oracle path (env + agent): exact
+agreement on decisions, delays, metrics, losses and gradients at B=1; batch
+linearity at B=2."""
+import numpy as np
+import pytest
+import torch
+
+from multihop_offload_amd import ACOAgent, AdhocCloudEnv, CaseGraph, JobInstance
+from multihop_offload_amd.agent import AgentConfig
+from multihop_offload_amd.engine import EpisodeEngine, JobBatch
+from multihop_offload_amd.env import apsp, delay_empirical
+
+
+def _case(seed=7, n=20):
+    rng = np.random.RandomState(42 + seed)
+    g = CaseGraph(n, t_max=1000, seed=seed, gtype="ba")
+    g.links_init(50.0, rng=rng)
+    g.add_relay(0)
+    g.add_relay(1)
+    for s in (2, 3, 4):
+        g.add_server(s, 300.0)
+    for v in range(5, n):
+        g.set_mobile_bw(v, 10.0)
+    return g
+
+
+def _jobbatch_from(engine, instances):
+    """Pack per-case JobInstance lists into a padded JobBatch."""
+    B, J = engine.B, engine.Jmax
+    src = np.zeros((B, J), dtype=np.int64)
+    mask = np.zeros((B, J), dtype=bool)
+    rates = np.zeros((B, J))
+    ul = np.zeros((B, J))
+    dl = np.zeros((B, J))
+    for b, jobs in enumerate(instances):
+        k = jobs.num_jobs
+        src[b, :k] = jobs.sources
+        src[b, k:] = jobs.sources[0] if k else 0
+        mask[b, :k] = True
+        rates[b, :k] = jobs.rates
+        ul[b, :k] = jobs.ul
+        dl[b, :k] = jobs.dl
+    dev, dt = engine.device, engine.dtype
+    return JobBatch(
+        sources=torch.as_tensor(src, device=dev),
+        mask=torch.as_tensor(mask, device=dev),
+        rates=torch.as_tensor(rates, dtype=dt, device=dev),
+        ul=torch.as_tensor(ul, dtype=dt, device=dev),
+        dl=torch.as_tensor(dl, dtype=dt, device=dev),
+    )
+
+
+def _wake(model):
+    with torch.no_grad():
+        for layer in model.layers:
+            layer.weight.mul_(0.01)
+        model.layers[-1].bias.fill_(0.5)
+
+
+@pytest.fixture
+def setup():
+    g = _case()
+    agent = ACOAgent(AgentConfig(T=1000, seed=5), 100)
+    _wake(agent.model)
+    engine = EpisodeEngine([g], agent.model, device="cpu",
+                           dtype=torch.float64)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, np.random.RandomState(3))
+    return g, agent, engine, jobs
+
+
+def test_gnn_episode_matches_oracle(setup):
+    g, agent, engine, jobs = setup
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    out = agent.forward_backward(env, 0.0, np.random.RandomState(0))
+    tau_oracle = np.nanmean(delay_empirical(out[1], out[2]))
+    grads_oracle = agent.memory[-1][0]
+    loss_fn_oracle, loss_mse_oracle = out[5], out[6]
+
+    jb = _jobbatch_from(engine, [jobs])
+    for p in engine.model.parameters():
+        p.grad = None
+    res = engine.gnn_episode(jb, explore=0.0, train=True)
+
+    # decisions/evaluation agree
+    emp = res.delay_emp[0, :jobs.num_jobs].numpy()
+    emp_oracle = delay_empirical(out[1], out[2])
+    assert np.allclose(emp, emp_oracle, rtol=1e-10, equal_nan=True)
+    assert np.isclose(res.tau[0].item(), tau_oracle)
+    assert res.congest[0].item() == np.count_nonzero(
+        emp_oracle > float(g.T))
+
+    # losses agree
+    assert np.isclose(res.loss_fn, loss_fn_oracle, rtol=1e-10)
+    assert np.isclose(res.loss_mse, loss_mse_oracle, rtol=1e-8)
+
+    # gradients agree
+    for p, go in zip(engine.model.parameters(), grads_oracle):
+        assert np.allclose(p.grad.numpy(), go.numpy(), rtol=1e-8, atol=1e-12)
+
+
+def test_baseline_and_local_match_oracle(setup):
+    g, agent, engine, jobs = setup
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+
+    dmtx, dlist, dproc = env.dmtx_baseline()
+    dproc2 = np.where(dproc > 0, dproc, float(g.T))
+    sp = apsp(g, dlist)
+    np.fill_diagonal(sp, dproc2)
+    env.offloading(sp, g.sp_hop)
+    ldel, sdel, _ = env.run()
+    tau_bl = np.nanmean(delay_empirical(ldel, sdel))
+
+    jb = _jobbatch_from(engine, [jobs])
+    res = engine.baseline_episode(jb)
+    assert np.isclose(res.tau[0].item(), tau_bl, rtol=1e-10)
+
+    env.set_jobs(jobs)
+    env.local_compute(dproc)
+    ldel, sdel, _ = env.run()
+    tau_lo = np.nanmean(delay_empirical(ldel, sdel))
+    res = engine.local_episode(jb)
+    assert np.isclose(res.tau[0].item(), tau_lo, rtol=1e-10)
+
+
+def test_batch_gradients_are_sum_of_instances():
+    g1, g2 = _case(seed=11), _case(seed=13)
+    agent = ACOAgent(AgentConfig(T=1000, seed=5), 100)
+    _wake(agent.model)
+    j1 = JobInstance.sample(g1.mobile_nodes, 0.15, np.random.RandomState(1))
+    j2 = JobInstance.sample(g2.mobile_nodes, 0.15, np.random.RandomState(2))
+
+    # per-instance gradients via the oracle path
+    for g, j in ((g1, j1), (g2, j2)):
+        env = AdhocCloudEnv(g)
+        env.set_jobs(j)
+        agent.forward_backward(env, 0.0, np.random.RandomState(0))
+    ga = agent.memory[-2][0]
+    gb = agent.memory[-1][0]
+
+    engine = EpisodeEngine([g1, g2], agent.model, device="cpu",
+                           dtype=torch.float64)
+    jb = _jobbatch_from(engine, [j1, j2])
+    for p in engine.model.parameters():
+        p.grad = None
+    res = engine.gnn_episode(jb, train=True)
+    assert res.tau.shape == (2,)
+    for p, a, b in zip(engine.model.parameters(), ga, gb):
+        assert np.allclose(p.grad.numpy(), (a + b).numpy(), rtol=1e-8,
+                           atol=1e-12)
+
+
+def test_sample_jobs_batch_properties():
+    g = _case()
+    agent = ACOAgent(AgentConfig(T=1000, seed=5), 10)
+    engine = EpisodeEngine([g, _case(seed=9)], agent.model, device="cpu",
+                           dtype=torch.float64)
+    gen = torch.Generator().manual_seed(0)
+    jb = engine.sample_jobs(0.15, gen)
+    assert jb.sources.shape == (2, engine.Jmax)
+    nj = jb.mask.sum(1)
+    M = engine.mobile_mask.sum(1)
+    assert ((nj >= (0.3 * M).to(torch.int64)) & (nj < M)).all()
+    # sources are mobiles, unique among real jobs
+    for b in range(2):
+        srcs = jb.sources[b][jb.mask[b]].numpy()
+        assert len(set(srcs.tolist())) == len(srcs)
+        assert all(engine.mobile_mask[b, s] for s in srcs)
+    assert (jb.rates[jb.mask] >= 0.1 * 0.15 - 1e-12).all()
+    assert (jb.rates[jb.mask] <= 0.5 * 0.15 + 1e-12).all()
+    assert (jb.rates[~jb.mask] == 0).all()
+
+
+def test_floyd_warshall_vs_scipy():
+    from multihop_offload_amd.ops import torch_ref
+    g = _case()
+    rng = np.random.RandomState(0)
+    w = rng.uniform(0.01, 2.0, g.num_links)
+    want = apsp(g, w)
+    N = g.num_nodes
+    wm = np.full((N, N), np.inf)
+    wm[g.edges[:, 0], g.edges[:, 1]] = w
+    wm[g.edges[:, 1], g.edges[:, 0]] = w
+    np.fill_diagonal(wm, 0)
+    got = torch_ref.floyd_warshall(
+        torch.tensor(wm, dtype=torch.float64)[None])[0].numpy()
+    assert np.allclose(got, want)

@@ -1,0 +1,151 @@
+"""GPU (MI355X) tests: HIP kernels vs torch references, end-to-end engine."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+needs_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                               reason="no GPU")
+
+
+@needs_gpu
+def test_hip_ext_loads():
+    from multihop_offload_amd.ops import dispatch
+    assert dispatch.hip_available(), \
+        "HIP extension must be built in-tree (setup.py build_ext --inplace)"
+
+
+def _rand_w(B, N, deg=4, seed=0):
+    rng = np.random.RandomState(seed)
+    w = np.full((B, N, N), np.inf, dtype=np.float32)
+    for b in range(B):
+        for _ in range(deg * N):
+            i, j = rng.randint(N), rng.randint(N)
+            if i != j:
+                v = rng.uniform(0.01, 2.0)
+                w[b, i, j] = w[b, j, i] = min(w[b, i, j], v)
+        # ensure connectivity via a ring
+        for i in range(N):
+            j = (i + 1) % N
+            v = rng.uniform(0.01, 2.0)
+            w[b, i, j] = w[b, j, i] = min(w[b, i, j], v)
+        np.fill_diagonal(w[b], 0.0)
+    return torch.tensor(w)
+
+
+@needs_gpu
+@pytest.mark.parametrize("N", [20, 100, 199])
+def test_fw_lds_kernel_vs_ref(N):
+    from multihop_offload_amd.ops import dispatch, torch_ref
+    w = _rand_w(8, N)
+    want = torch_ref.floyd_warshall(w.double())
+    got = dispatch.floyd_warshall(w.cuda()).cpu()
+    assert torch.allclose(got.double(), want, rtol=1e-5, atol=1e-5)
+
+
+@needs_gpu
+@pytest.mark.parametrize("N", [256, 513])
+def test_fw_tiled_kernel_vs_ref(N):
+    from multihop_offload_amd.ops import dispatch, torch_ref
+    w = _rand_w(2, N, deg=3)
+    want = torch_ref.floyd_warshall(w.double())
+    got = dispatch.floyd_warshall(w.cuda()).cpu()
+    assert torch.allclose(got.double(), want, rtol=1e-4, atol=1e-4)
+
+
+def _cases(n=20, B=8):
+    from multihop_offload_amd.graphs import CaseGraph
+    rng = np.random.RandomState(0)
+    cases = []
+    for b in range(B):
+        g = CaseGraph(n, t_max=1000, seed=b + 1, gtype="ba")
+        g.links_init(50.0, rng=rng)
+        g.add_relay(0)
+        for s in (2, 3, 4):
+            g.add_server(s, 300.0)
+        for v in range(5, n):
+            g.set_mobile_bw(v, 10.0)
+        cases.append(g)
+    return cases
+
+
+@needs_gpu
+def test_engine_gpu_matches_cpu():
+    """Same fp32 model, same injected jobs: GPU and CPU engines must agree
+    on metrics and gradients (tolerances cover reduction-order effects)."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from tests.test_engine import _jobbatch_from
+    from multihop_offload_amd.graphs import JobInstance
+
+    cases = _cases()
+    model_c = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    model_g = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    with torch.no_grad():
+        for pc, pg in zip(model_c.parameters(), model_g.parameters()):
+            pc.mul_(0.01)
+            pg.copy_(pc)
+        model_c.layers[-1].bias.fill_(0.5)
+        model_g.layers[-1].bias.fill_(0.5)
+
+    eng_c = EpisodeEngine(cases, model_c, device="cpu", dtype=torch.float32)
+    eng_g = EpisodeEngine(cases, model_g, device="cuda", dtype=torch.float32)
+    insts = [JobInstance.sample(c.mobile_nodes, 0.15,
+                                np.random.RandomState(10 + i))
+             for i, c in enumerate(cases)]
+    jb_c = _jobbatch_from(eng_c, insts)
+    jb_g = _jobbatch_from(eng_g, insts)
+
+    res_c = eng_c.gnn_episode(jb_c, train=True)
+    res_g = eng_g.gnn_episode(jb_g, train=True)
+    tau_c, tau_g = res_c.tau.numpy(), res_g.tau.cpu().numpy()
+    assert np.allclose(tau_c, tau_g, rtol=1e-3)
+    assert np.isclose(res_c.loss_fn, res_g.loss_fn, rtol=1e-3)
+    for pc, pg in zip(model_c.parameters(), model_g.parameters()):
+        a, b = pc.grad.numpy(), pg.grad.cpu().numpy()
+        denom = max(np.abs(a).max(), 1e-6)
+        assert np.abs(a - b).max() / denom < 5e-3
+
+
+@needs_gpu
+def test_engine_gpu_train_step_moves_params():
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    cases = _cases()
+    model = ChebConvStack(K=2, dtype=torch.float32, seed=0)
+    with torch.no_grad():
+        for layer in model.layers:
+            layer.weight.mul_(0.01)
+        model.layers[-1].bias.fill_(0.5)
+    engine = EpisodeEngine(cases, model, device="cuda", dtype=torch.float32)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3, eps=1e-7)
+    gen = torch.Generator(device="cuda")
+    gen.manual_seed(0)
+    before = [p.detach().clone() for p in model.parameters()]
+    for _ in range(2):
+        jobs = engine.sample_jobs(0.15, gen)
+        for p in model.parameters():
+            p.grad = None
+        res = engine.gnn_episode(jobs, train=True, gen=gen)
+        opt.step()
+        model.apply_constraints()
+    assert torch.isfinite(res.tau).all()
+    assert any(not torch.equal(b, p.detach())
+               for b, p in zip(before, model.parameters()))
+
+
+@needs_gpu
+def test_baseline_local_episodes_gpu():
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    cases = _cases()
+    model = ChebConvStack(K=2, dtype=torch.float32, seed=0)
+    engine = EpisodeEngine(cases, model, device="cuda", dtype=torch.float32)
+    gen = torch.Generator(device="cuda")
+    gen.manual_seed(1)
+    jobs = engine.sample_jobs(0.15, gen)
+    rb = engine.baseline_episode(jobs)
+    rl = engine.local_episode(jobs)
+    assert torch.isfinite(rb.tau).all() and torch.isfinite(rl.tau).all()
+    assert (rb.num_jobs == rl.num_jobs).all()
