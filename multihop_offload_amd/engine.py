@@ -177,6 +177,33 @@ class EpisodeEngine:
 
         self._bidx = torch.arange(B, device=self.device)
 
+        # ---- kernel tables (int32 CSR forms for the fused HIP kernels) ----
+        def t32(a):
+            return torch.as_tensor(np.asarray(a, dtype=np.int32),
+                                   device=self.device)
+
+        self.k_adj_indptr = t32([c.adj_indptr for c in cases])     # (B,N+1)
+        self.k_adj_idx = t32([c.adj_indices for c in cases])       # (B,2E)
+        self.k_adj_link = t32([c.adj_link_ids for c in cases])     # (B,2E)
+        self.k_conf_indptr = t32([c.conf_indptr for c in cases])   # (B,E+1)
+        cc, base = [], [0]
+        for c in cases:
+            cc.append(np.asarray(c.conf_indices, dtype=np.int32))
+            base.append(base[-1] + len(c.conf_indices))
+        self.k_conf_cols = t32(np.concatenate(cc))
+        self.k_conf_base = torch.as_tensor(np.asarray(base[:-1]),
+                                           dtype=torch.int64,
+                                           device=self.device)
+        self.k_edges = t32([c.edges for c in cases])               # (B,E,2)
+        self.k_servers = t32(srv)                                  # (B,S)
+        import os
+        self.use_hip = (self.device.type == "cuda"
+                        and os.environ.get("MHO_FORCE_TORCH") != "1")
+        if self.use_hip:
+            assert dtype == torch.float32, "HIP kernels are fp32"
+            from .ops import dispatch
+            dispatch.require_hip()
+
     # ------------------------------------------------------------------ jobs
     def sample_jobs(self, arrival_scale: float,
                     gen: Optional[torch.Generator] = None) -> JobBatch:
@@ -227,6 +254,11 @@ class EpisodeEngine:
         x = torch.stack([self.f_self_loop, self.f_rate, f_job,
                          self.f_as_server], dim=-1).reshape(B * Ee, 4)
         lam = self.model(x, self.support).reshape(B, Ee)
+
+        if self.use_hip:
+            from .ops.functions import ActorHeadFn
+            dm = ActorHeadFn.apply(lam, self)
+            return dm, None, None
 
         lam_link = lam[:, :E].reshape(-1)
         mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),
@@ -282,6 +314,24 @@ class EpisodeEngine:
         ``sp``: (B,N,N) shortest-path delays with ZERO diagonal;
         ``uds``: (B,N) per-node unit processing delays (inf at relays)."""
         B, J, S = self.B, self.Jmax, self.S
+        if self.use_hip and not prob:
+            from .ops import dispatch
+            ext = dispatch.require_hip()
+            dst, islocal = ext.decide(
+                sp.contiguous(), self.sp_hop.contiguous(), uds.contiguous(),
+                self.k_servers, jobs.sources, jobs.mask,
+                jobs.ul.contiguous(), jobs.dl.contiguous())
+            if explore > 0:
+                nS = self.server_mask.sum(1, keepdim=True)
+                r = torch.rand(B, J, device=self.device, generator=gen)
+                rc = (torch.rand(B, J, device=self.device, generator=gen)
+                      * (nS + 1).to(self.dtype)).to(torch.int64)
+                dst_rand = torch.where(
+                    rc >= nS, jobs.sources,
+                    self.servers_safe.gather(1, rc.clamp(max=S - 1)))
+                dst = torch.where(jobs.mask & (r < explore), dst_rand, dst)
+            return dst, None
+
         bJ = self._bidx[:, None]
         src = jobs.sources                                    # (B,J)
         local = uds.gather(1, src) * jobs.ul                  # (B,J)
@@ -368,6 +418,28 @@ class EpisodeEngine:
         return route_links, hops
 
     # --------------------------------------------------------- evaluation
+    def _episode_eval(self, jobs: JobBatch, dst: torch.Tensor,
+                      sp: torch.Tensor):
+        """Walk + evaluate: fused HIP kernel on GPU, torch path on CPU.
+        Returns (route_links, nhop, delay_emp, unit_mtx, written)."""
+        if self.use_hip:
+            from .ops import dispatch
+            ext = dispatch.require_hip()
+            H = min(self.walk_cap, 64)
+            rl, nhop, delay_emp, unit_mtx, written, overflow = ext.walk_eval(
+                sp.contiguous(), jobs.sources, dst.contiguous(), jobs.mask,
+                jobs.rates.contiguous(), jobs.ul.contiguous(),
+                jobs.dl.contiguous(), self.k_adj_indptr, self.k_adj_idx,
+                self.k_adj_link, self.k_conf_indptr, self.k_conf_base,
+                self.k_conf_cols, self.link_rates.contiguous(),
+                self.proc_bws.contiguous(), self.k_edges,
+                float(self.T), H, self.fp_iters)
+            self._last_overflow = overflow
+            return rl, nhop, delay_emp, unit_mtx, written
+        rl, nhop = self.route_walk(jobs, dst, sp)
+        delay_emp, unit_mtx, written, *_ = self.evaluate(jobs, dst, rl, nhop)
+        return rl, nhop, delay_emp, unit_mtx, written
+
     def evaluate(self, jobs: JobBatch, dst: torch.Tensor,
                  route_links: torch.Tensor, nhop: torch.Tensor):
         """Analytic queueing evaluation (offloading_v3.py:455-550), batched.
@@ -451,16 +523,28 @@ class EpisodeEngine:
 
     # ------------------------------------------------------------- critic
     def critic_backward(self, jobs: JobBatch, dst: torch.Tensor,
-                        route_links: torch.Tensor):
+                        route_links: torch.Tensor,
+                        nhop: Optional[torch.Tensor] = None):
         """Critic loss over the routes tensor + grad wrt routes
         (gnn_offloading_agent.py:333-374) and the closed-form route-bias
         cotangent accumulation (:384-416), batched.  Returns
         (grad_edge (B,Ē), loss_fn scalar)."""
         B, E, Ee, J = self.B, self.E, self.Ee, self.Jmax
+        vedge_dst = self.node_vedge.gather(1, dst)            # (B,J)
+        if self.use_hip:
+            from .ops import dispatch
+            ext = dispatch.require_hip()
+            grad_edge, loss = ext.critic(
+                route_links.contiguous(), nhop.contiguous(),
+                vedge_dst.contiguous(), jobs.mask,
+                jobs.rates.contiguous(), jobs.ul.contiguous(),
+                jobs.dl.contiguous(), self.k_conf_indptr, self.k_conf_base,
+                self.k_conf_cols, self.link_rates.contiguous(),
+                self.bw_comp.contiguous(), float(self.T), Ee, self.fp_iters)
+            return grad_edge, float(loss.sum())
         H = route_links.shape[2]
         valid = route_links >= 0
         safe = route_links.clamp(min=0)
-        vedge_dst = self.node_vedge.gather(1, dst)            # (B,J)
 
         # routes (B,Ē,J) — link hops + destination self-loop edge
         routes = torch.zeros(B, Ee, J, dtype=self.dtype, device=self.device)
@@ -554,13 +638,13 @@ class EpisodeEngine:
         idx = torch.arange(self.N, device=self.device)
         uds = dm.detach()[:, idx, idx]                        # (B,N)
         dst, est = self.offload_decide(jobs, sp, uds, explore, gen, prob)
-        route_links, nhop = self.route_walk(jobs, dst, sp)
-        delay_emp, unit_mtx, written, lam_r, mu_r, server_load = \
-            self.evaluate(jobs, dst, route_links, nhop)
+        route_links, nhop, delay_emp, unit_mtx, written = \
+            self._episode_eval(jobs, dst, sp)
 
         loss_fn = loss_mse = None
         if train:
-            grad_edge, loss_fn = self.critic_backward(jobs, dst, route_links)
+            grad_edge, loss_fn = self.critic_backward(jobs, dst, route_links,
+                                                      nhop)
             grad_dist, loss_mse = self.grad_dist_matrix(
                 grad_edge, dm, unit_mtx, written)
             dm.backward(grad_dist)
@@ -587,8 +671,7 @@ class EpisodeEngine:
             uds = torch.where((self.proc_bws > 0),
                               dproc, torch.full_like(dproc, float("inf")))
             dst, est = self.offload_decide(jobs, sp, uds)
-            route_links, nhop = self.route_walk(jobs, dst, sp)
-            delay_emp, *_ = self.evaluate(jobs, dst, route_links, nhop)
+            _, _, delay_emp, _, _ = self._episode_eval(jobs, dst, sp)
             nj = jobs.mask.sum(1)
             de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
             return EpisodeResult(
@@ -601,10 +684,17 @@ class EpisodeEngine:
         B, J = self.B, self.Jmax
         with torch.no_grad():
             dst = jobs.sources
-            route_links = torch.full((B, J, 0), -1, dtype=torch.int64,
-                                     device=self.device)
-            nhop = torch.zeros(B, J, dtype=torch.int64, device=self.device)
-            delay_emp, *_ = self.evaluate(jobs, dst, route_links, nhop)
+            if self.use_hip:
+                # the fused kernel handles src==dst routes (0 hops) directly
+                sp0 = torch.zeros(B, self.N, self.N, dtype=self.dtype,
+                                  device=self.device)
+                _, _, delay_emp, _, _ = self._episode_eval(jobs, dst, sp0)
+            else:
+                route_links = torch.full((B, J, 0), -1, dtype=torch.int64,
+                                         device=self.device)
+                nhop = torch.zeros(B, J, dtype=torch.int64,
+                                   device=self.device)
+                delay_emp, *_ = self.evaluate(jobs, dst, route_links, nhop)
             nj = jobs.mask.sum(1)
             de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
             return EpisodeResult(

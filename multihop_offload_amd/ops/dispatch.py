@@ -35,7 +35,7 @@ def hip_available() -> bool:
     return _load_hip() is not None
 
 
-def _require_hip():
+def require_hip():
     ext = _load_hip()
     if ext is None:
         raise RuntimeError(
@@ -43,6 +43,9 @@ def _require_hip():
             "`python setup.py build_ext --inplace` (gfx950). GPU execution "
             "without the native kernels is disabled by design.")
     return ext
+
+
+_require_hip = require_hip
 
 
 def floyd_warshall(w: torch.Tensor) -> torch.Tensor:

@@ -2,8 +2,39 @@
 #include <torch/extension.h>
 
 torch::Tensor floyd_warshall_hip(torch::Tensor w);
+std::vector<torch::Tensor> decide_hip(
+    torch::Tensor sp, torch::Tensor hop, torch::Tensor uds,
+    torch::Tensor servers, torch::Tensor src, torch::Tensor mask,
+    torch::Tensor ul, torch::Tensor dl);
+std::vector<torch::Tensor> walk_eval_hip(
+    torch::Tensor sp, torch::Tensor src, torch::Tensor dst,
+    torch::Tensor mask, torch::Tensor rate, torch::Tensor ul,
+    torch::Tensor dl, torch::Tensor adj_indptr, torch::Tensor adj_idx,
+    torch::Tensor adj_link, torch::Tensor conf_indptr,
+    torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
+    torch::Tensor bw, torch::Tensor edges, double T, long H, long fp_iters);
+std::vector<torch::Tensor> critic_hip(
+    torch::Tensor route_links, torch::Tensor nhop, torch::Tensor vedge_dst,
+    torch::Tensor mask, torch::Tensor rate, torch::Tensor ul,
+    torch::Tensor dl, torch::Tensor conf_indptr, torch::Tensor conf_base,
+    torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
+    double T, long Ee, long iters);
+std::vector<torch::Tensor> actor_head_fwd_hip(
+    torch::Tensor lam_ext, torch::Tensor conf_indptr,
+    torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
+    torch::Tensor bw_comp, torch::Tensor edges, torch::Tensor node_vedge,
+    double T, long N, long iters);
+torch::Tensor actor_head_bwd_hip(
+    torch::Tensor grad_dist, torch::Tensor lam_ext, torch::Tensor mu_hist,
+    torch::Tensor conf_indptr, torch::Tensor conf_base,
+    torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
+    torch::Tensor edges, torch::Tensor node_vedge, double T, long iters);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-    m.def("floyd_warshall", &floyd_warshall_hip,
-          "Batched min-plus Floyd-Warshall APSP (gfx950)");
+    m.def("floyd_warshall", &floyd_warshall_hip);
+    m.def("decide", &decide_hip);
+    m.def("walk_eval", &walk_eval_hip);
+    m.def("critic", &critic_hip);
+    m.def("actor_head_fwd", &actor_head_fwd_hip);
+    m.def("actor_head_bwd", &actor_head_bwd_hip);
 }

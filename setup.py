@@ -19,6 +19,8 @@ setup(
             sources=[
                 "multihop_offload_amd/ops/hip/bindings.cpp",
                 "multihop_offload_amd/ops/hip/fw.hip",
+                "multihop_offload_amd/ops/hip/episode.hip",
+                "multihop_offload_amd/ops/hip/queueing.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3"],
