@@ -196,6 +196,21 @@ class EpisodeEngine:
                                            device=self.device)
         self.k_edges = t32([c.edges for c in cases])               # (B,E,2)
         self.k_servers = t32(srv)                                  # (B,S)
+        # extended line-graph CSR (ChebConv support), padded rows empty
+        ipt = np.zeros((B, Ee + 1), dtype=np.int32)
+        cc, base = [], [0]
+        for b, c in enumerate(cases):
+            ext = c.ext
+            ne = ext.num_edges_ext
+            ipt[b, :ne + 1] = ext.ext_indptr
+            ipt[b, ne + 1:] = ext.ext_indptr[-1]
+            cc.append(np.asarray(ext.ext_indices, dtype=np.int32))
+            base.append(base[-1] + len(ext.ext_indices))
+        self.k_ext_indptr = t32(ipt)
+        self.k_ext_cols = t32(np.concatenate(cc))
+        self.k_ext_base = torch.as_tensor(np.asarray(base[:-1]),
+                                          dtype=torch.int64,
+                                          device=self.device)
         import os
         self.use_hip = (self.device.type == "cuda"
                         and os.environ.get("MHO_FORCE_TORCH") != "1")
@@ -252,13 +267,22 @@ class EpisodeEngine:
             1, vidx, torch.where(self.comp_mask, arr, torch.zeros_like(arr)))
 
         x = torch.stack([self.f_self_loop, self.f_rate, f_job,
-                         self.f_as_server], dim=-1).reshape(B * Ee, 4)
-        lam = self.model(x, self.support).reshape(B, Ee)
+                         self.f_as_server], dim=-1)               # (B,Ē,4)
 
         if self.use_hip:
-            from .ops.functions import ActorHeadFn
+            from .ops.functions import ActorHeadFn, ChebStackFn, cheb_compat
+            if cheb_compat(self.model):
+                params = []
+                for layer in self.model.layers:
+                    params += [layer.weight, layer.bias]
+                lam = ChebStackFn.apply(x.contiguous(), self, *params)
+            else:  # unsupported model shape: torch layers on GPU
+                lam = self.model(x.reshape(B * Ee, 4),
+                                 self.support).reshape(B, Ee)
             dm = ActorHeadFn.apply(lam, self)
             return dm, None, None
+
+        lam = self.model(x.reshape(B * Ee, 4), self.support).reshape(B, Ee)
 
         lam_link = lam[:, :E].reshape(-1)
         mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),

@@ -30,6 +30,13 @@ torch::Tensor actor_head_bwd_hip(
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
     torch::Tensor edges, torch::Tensor node_vedge, double T, long iters);
 
+std::vector<torch::Tensor> cheb_fwd_hip(
+    torch::Tensor x, torch::Tensor W, torch::Tensor bias,
+    torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols);
+std::vector<torch::Tensor> cheb_bwd_hip(
+    torch::Tensor dlam, torch::Tensor acts, torch::Tensor W,
+    torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("floyd_warshall", &floyd_warshall_hip);
     m.def("decide", &decide_hip);
@@ -37,4 +44,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("critic", &critic_hip);
     m.def("actor_head_fwd", &actor_head_fwd_hip);
     m.def("actor_head_bwd", &actor_head_bwd_hip);
+    m.def("cheb_fwd", &cheb_fwd_hip);
+    m.def("cheb_bwd", &cheb_bwd_hip);
 }

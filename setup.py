@@ -21,6 +21,7 @@ setup(
                 "multihop_offload_amd/ops/hip/fw.hip",
                 "multihop_offload_amd/ops/hip/episode.hip",
                 "multihop_offload_amd/ops/hip/queueing.hip",
+                "multihop_offload_amd/ops/hip/chebconv.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3"],
