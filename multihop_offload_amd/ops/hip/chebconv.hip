@@ -37,10 +37,11 @@ DEV_INLINE void gemm_acc(const float* __restrict__ src,
                          int rows_pad, int tid) {
     const int lane = tid & 63;
     const int wid = tid >> 6;
+    const int nw = blockDim.x >> 6;
     const int mtiles = rows_pad / 16;
     const int r_in = lane & 15;          // A-operand row within tile
     const int k_in = lane >> 4;          // A/B k index (0..3)
-    for (int t = wid; t < mtiles * 2; t += 4) {
+    for (int t = wid; t < mtiles * 2; t += nw) {
         const int mt = t >> 1;
         const int c0 = (t & 1) * 16;
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -58,6 +59,32 @@ DEV_INLINE void gemm_acc(const float* __restrict__ src,
             const int row = mt * 16 + (lane >> 4) * 4 + r;
             dst[row * STRIDE + c0 + (lane & 15)] += acc[r];
         }
+    }
+}
+
+// ---- weight-gradient tile: dW[i][j] = sum_r src[r][i] * delta[r][j] ------
+// MFMA over the row (K) dimension: 2×2 tiles of 16×16, one per wave.
+DEV_INLINE void gemm_wgrad(const float* __restrict__ src,
+                           const float* __restrict__ delta,
+                           float* __restrict__ dw_out,  // global [32][32]
+                           int rows_pad, int tid) {
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int k_in = lane >> 4;
+    const int c_in = lane & 15;
+    for (int t = wid; t < 4; t += (blockDim.x >> 6)) {
+        const int i0 = (t >> 1) * 16, j0 = (t & 1) * 16;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        for (int kk = 0; kk < rows_pad / 4; ++kk) {
+            const int r = kk * 4 + k_in;
+            const float a = src[r * STRIDE + i0 + c_in];
+            const float b = delta[r * STRIDE + j0 + c_in];
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            dw_out[(i0 + (lane >> 4) * 4 + r) * F + j0 + (lane & 15)] =
+                acc[r];
     }
 }
 
@@ -235,31 +262,21 @@ __global__ void cheb_bwd_kernel(
             Wl[i] = W[((size_t)l * K) * F * F + i];
         __syncthreads();
 
-        // db[j] = sum_r Db[r][j]
-        for (int j = tid; j < F; j += nt) {
+        // db[j] = sum_r Db[r][j] — (j, row-chunk) threads, db prezeroed
+        {
+            const int nchunk = nt / F;
+            const int j = tid % F, ch = tid / F;
             float acc = 0.f;
-            for (int r = 0; r < Ee; ++r) acc += Db[r * STRIDE + j];
-            dbb[l * F + j] = acc;
+            for (int r = ch; r < Ee; r += nchunk) acc += Db[r * STRIDE + j];
+            atomicAdd(&dbb[l * F + j], acc);
         }
-        // dW0[i][j] = sum_r Ab[r][i] * Db[r][j]
-        for (int p = tid; p < F * F; p += nt) {
-            const int i = p / F, j = p % F;
-            float acc = 0.f;
-            for (int r = 0; r < Ee; ++r)
-                acc += Ab[r * STRIDE + i] * Db[r * STRIDE + j];
-            dWb[((size_t)l * K) * F * F + p] = acc;
-        }
+        gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
         if (K > 1) {
             __syncthreads();
             spmv(Ab, Tb, ipt, cls, Ee, rows_pad, tid, nt, 0);  // T1 = A·X_l
             __syncthreads();
-            for (int p = tid; p < F * F; p += nt) {
-                const int i = p / F, j = p % F;
-                float acc = 0.f;
-                for (int r = 0; r < Ee; ++r)
-                    acc += Tb[r * STRIDE + i] * Db[r * STRIDE + j];
-                dWb[((size_t)l * K + 1) * F * F + p] = acc;
-            }
+            gemm_wgrad(Tb, Db, dWb + ((size_t)l * K + 1) * F * F, rows_pad,
+                       tid);
         }
         if (l == 0) break;                       // features are leaves
         __syncthreads();
@@ -303,7 +320,7 @@ std::vector<torch::Tensor> cheb_fwd_hip(
         (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F + F);
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for fused ChebConv");
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(cheb_fwd_kernel, dim3(B), dim3(256), lds,
+    hipLaunchKernelGGL(cheb_fwd_kernel, dim3(B), dim3(512), lds,
                        stream.stream(),
                        x.data_ptr<float>(), W.data_ptr<float>(),
                        bias.data_ptr<float>(), ext_indptr.data_ptr<int>(),
@@ -326,7 +343,7 @@ std::vector<torch::Tensor> cheb_bwd_hip(
         (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F);
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for fused ChebConv bwd");
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(cheb_bwd_kernel, dim3(B), dim3(256), lds,
+    hipLaunchKernelGGL(cheb_bwd_kernel, dim3(B), dim3(512), lds,
                        stream.stream(),
                        dlam.data_ptr<float>(), acts.data_ptr<float>(),
                        W.data_ptr<float>(), ext_indptr.data_ptr<int>(),

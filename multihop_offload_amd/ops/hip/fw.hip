@@ -32,12 +32,18 @@ __global__ void fw_lds_kernel(T* __restrict__ d, int N) {
     const int total = N * N;
     for (int c = tid; c < total; c += nt) s[c] = D[c];
     __syncthreads();
+    // thread layout: column j fixed per thread (nc columns per stripe of
+    // 128 lanes), rows strided — d[k][j] is loaded once per k and d[i][k]
+    // broadcasts across the j-lanes of a wave.
+    const int i0 = tid >> 7;
+    const int istep = nt >> 7;
     for (int k = 0; k < N; ++k) {
-        for (int c = tid; c < total; c += nt) {
-            const int i = c / N;
-            const int j = c - i * N;
-            const T alt = s[i * N + k] + s[k * N + j];
-            if (alt < s[c]) s[c] = alt;
+        for (int j = tid & 127; j < N; j += 128) {
+            const T dkj = s[k * N + j];
+            for (int i = i0; i < N; i += istep) {
+                const T alt = s[i * N + k] + dkj;
+                if (alt < s[i * N + j]) s[i * N + j] = alt;
+            }
         }
         __syncthreads();
     }
