@@ -28,37 +28,15 @@ import numpy as np
 import torch
 
 
-def build_cases(n_nodes, batch, distinct, T, seed):
-    """Synthetic cases of the named config: BA(m=2) topologies with the
-    datagen role/bandwidth distributions; the batch replicates `distinct`
+def build_cases(n_nodes, batch, distinct, T, seed, gtype="ba", workers=8):
+    """Synthetic cases of the named config: topologies with the datagen
+    role/bandwidth distributions; the batch replicates `distinct`
     topologies with independent link-rate draws (data=synthetic)."""
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
-    from multihop_offload_amd.datagen import generate_case
-    from multihop_offload_amd.graphs import CaseGraph
-
-    rng = np.random.RandomState(seed)
-    protos = []
-    for d in range(distinct):
-        case, _ = generate_case(n_nodes, seed + d, "ba", rng=rng)
-        protos.append(case)
-    cases = []
-    for b in range(batch):
-        case = protos[b % distinct]
-        adj = np.asarray(case["adj"].todense())
-        g = CaseGraph(n_nodes, t_max=T, seed=seed + b % distinct, gtype="ba",
-                      adj=adj, pos=case["pos_c"])
-        nodes_info = case["nodes_info"]
-        for nidx in range(n_nodes):
-            role, bw = nodes_info[nidx, 0], float(nodes_info[nidx, 1])
-            if role == 2:
-                g.add_relay(nidx)
-            elif role == 1:
-                g.add_server(nidx, bw)
-            else:
-                g.set_mobile_bw(nidx, bw)
-        g.links_init(case["link_rate"], rng=rng)
-        cases.append(g)
-    return cases
+    from multihop_offload_amd.harness.train_batched import \
+        build_training_cases
+    return build_training_cases(n_nodes, batch, distinct, T, seed,
+                                gtype=gtype, workers=workers)
 
 
 def main():

@@ -208,6 +208,26 @@ class CaseGraph:
         self.link_rates = np.round(np.clip(
             rng.normal(rates, std, size=(self.num_links,)), 0, hi))
 
+    def clone_with_rates(self, base_rates,
+                         rng: Optional[np.random.RandomState] = None,
+                         std: float = 2.0) -> "CaseGraph":
+        """Cheap replica: shares all topology structure (CSR, maps, hop
+        APSP), draws fresh link rates.  Used to build large batches from a
+        few distinct topologies without re-running graph construction."""
+        import copy as _copy
+        g = _copy.copy(self)
+        rng = rng or np.random
+        base = np.asarray(base_rates, dtype=np.float64)
+        g.link_rates = np.round(np.clip(
+            rng.normal(base, std, size=(self.num_links,)), 0,
+            base + 3 * std))
+        if self._ext is not None:
+            e = _copy.copy(self._ext)
+            e._g = g
+            e.edge_rate_ext = e.edge_rate_ext.copy()
+            g._ext = e
+        return g
+
     # -- conflict radius augmentation (offloading_v3.py:193-224) --------------
     def _add_conflict_relations(self, conf_sets):
         if self.pos is None:
