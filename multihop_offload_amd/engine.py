@@ -176,6 +176,9 @@ class EpisodeEngine:
         self._comp_r = ti(cr)
 
         self._bidx = torch.arange(B, device=self.device)
+        # per-graph horizon T (BASELINE config 5: ragged batches may mix T)
+        self.T_arr = t([float(c.T) for c in cases])            # (B,)
+        self.T_link = self.T_arr.repeat_interleave(E)          # (B*E,)
 
         # ---- kernel tables (int32 CSR forms for the fused HIP kernels) ----
         def t32(a):
@@ -288,10 +291,10 @@ class EpisodeEngine:
         mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),
                             self.cf_degs.reshape(-1), self.conf,
                             self.fp_iters)
-        link_delay = delay_with_fallback(lam_link, mu, self.T, 101.0) \
+        link_delay = delay_with_fallback(lam_link, mu, self.T_link, 101.0) \
             .reshape(B, E)
         node_delay = delay_with_fallback(lam[:, E:], self.bw_comp,
-                                         self.T, 100.0)       # (B, Ē-E)
+                                         self.T_arr[:, None], 100.0)
 
         dm = self._delay_matrix(link_delay, node_delay)
         return dm, link_delay, node_delay
@@ -457,7 +460,7 @@ class EpisodeEngine:
                 self.k_adj_link, self.k_conf_indptr, self.k_conf_base,
                 self.k_conf_cols, self.link_rates.contiguous(),
                 self.proc_bws.contiguous(), self.k_edges,
-                float(self.T), H, self.fp_iters)
+                self.T_arr.contiguous(), H, self.fp_iters)
             self._last_overflow = overflow
             return rl, nhop, delay_emp, unit_mtx, written
         rl, nhop = self.route_walk(jobs, dst, sp)
@@ -500,7 +503,7 @@ class EpisodeEngine:
         gap = mu_h - lam_h
         unit = torch.where(gap > 0, 1.0 / torch.where(gap > 0, gap,
                                                       torch.ones_like(gap)),
-                           float(self.T) * lam_h / (tot * mu_h))
+                           self.T_arr[:, None, None] * lam_h / (tot * mu_h))
         nh = nhop.to(self.dtype)[..., None]
         hop_delay = (torch.maximum(jobs.ul[..., None] * unit, nh)
                      + torch.maximum(jobs.dl[..., None] * unit, nh))
@@ -512,7 +515,7 @@ class EpisodeEngine:
         sgap = bw_dst - sl_dst
         sunit = torch.where(
             sgap > 0, 1.0 / torch.where(sgap > 0, sgap, torch.ones_like(sgap)),
-            float(self.T) * sl_dst / (jobs.ul * bw_dst))
+            self.T_arr[:, None] * sl_dst / (jobs.ul * bw_dst))
         server_part = torch.maximum(jobs.ul * sunit, torch.ones_like(sunit))
         delay_emp = torch.where(jobs.mask, link_part + server_part,
                                 torch.full_like(link_part, float("nan")))
@@ -564,7 +567,8 @@ class EpisodeEngine:
                 jobs.rates.contiguous(), jobs.ul.contiguous(),
                 jobs.dl.contiguous(), self.k_conf_indptr, self.k_conf_base,
                 self.k_conf_cols, self.link_rates.contiguous(),
-                self.bw_comp.contiguous(), float(self.T), Ee, self.fp_iters)
+                self.bw_comp.contiguous(), self.T_arr.contiguous(), Ee,
+                self.fp_iters)
             return grad_edge, float(loss.sum())
         H = route_links.shape[2]
         valid = route_links >= 0
@@ -593,10 +597,10 @@ class EpisodeEngine:
         mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),
                             self.cf_degs.reshape(-1), self.conf,
                             self.fp_iters)
-        link_d = delay_with_fallback(lam_link, mu, self.T, 101.0) \
+        link_d = delay_with_fallback(lam_link, mu, self.T_link, 101.0) \
             .reshape(B, E)
         node_d = delay_with_fallback(link_load[:, E:], self.bw_comp,
-                                     self.T, 100.0)
+                                     self.T_arr[:, None], 100.0)
         unit_edge = torch.cat([link_d, node_d], dim=1)         # (B,Ē)
         delay_job_edge = torch.maximum(
             jobs_data * unit_edge[..., None] * routes, routes)
@@ -676,7 +680,7 @@ class EpisodeEngine:
         nj = jobs.mask.sum(1)
         de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
         tau = de.sum(1) / nj.to(self.dtype)
-        congest = ((delay_emp > float(self.T)) & jobs.mask).sum(1)
+        congest = ((delay_emp > self.T_arr[:, None]) & jobs.mask).sum(1)
         return EpisodeResult(tau=tau, congest=congest, num_jobs=nj,
                              delay_emp=delay_emp, loss_fn=loss_fn,
                              loss_mse=loss_mse)
@@ -700,7 +704,7 @@ class EpisodeEngine:
             de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
             return EpisodeResult(
                 tau=de.sum(1) / nj.to(self.dtype),
-                congest=((delay_emp > float(self.T)) & jobs.mask).sum(1),
+                congest=((delay_emp > self.T_arr[:, None]) & jobs.mask).sum(1),
                 num_jobs=nj, delay_emp=delay_emp)
 
     def local_episode(self, jobs: JobBatch) -> EpisodeResult:
@@ -723,5 +727,5 @@ class EpisodeEngine:
             de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
             return EpisodeResult(
                 tau=de.sum(1) / nj.to(self.dtype),
-                congest=((delay_emp > float(self.T)) & jobs.mask).sum(1),
+                congest=((delay_emp > self.T_arr[:, None]) & jobs.mask).sum(1),
                 num_jobs=nj, delay_emp=delay_emp)

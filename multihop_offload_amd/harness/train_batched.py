@@ -100,6 +100,8 @@ def main(argv=None):
     ap.add_argument("--save_every", type=int, default=500)
     ap.add_argument("--log_every", type=int, default=50)
     ap.add_argument("--workers", type=int, default=8)
+    ap.add_argument("--lr_decay_at", type=int, default=0,
+                    help="step after which lr is multiplied by 0.1")
     ap.add_argument("--init_scale", type=float, default=0.01,
                     help="shrink initial weights (wakes the output ReLU)")
     args = ap.parse_args(argv)
@@ -156,6 +158,9 @@ def main(argv=None):
         opt.step()
         engine.model.apply_constraints()
         explore = max(explore * args.explore_decay, 0.001)
+        if args.lr_decay_at and step == args.lr_decay_at:
+            for group in opt.param_groups:
+                group["lr"] *= 0.1
 
         if step % args.log_every == 0 and rank == 0:
             tau = float(torch.nanmean(res.tau))

@@ -74,7 +74,7 @@ class ActorHeadFn(torch.autograd.Function):
             lam_ext.contiguous(), eng.k_conf_indptr, eng.k_conf_base,
             eng.k_conf_cols, eng.link_rates.contiguous(),
             eng.bw_comp.contiguous(), eng.k_edges, eng.node_vedge,
-            float(eng.T), eng.N, eng.fp_iters)
+            eng.T_arr.contiguous(), eng.N, eng.fp_iters)
         ctx.save_for_backward(lam_ext, mu_hist)
         ctx.eng = eng
         return dm
@@ -88,5 +88,6 @@ class ActorHeadFn(torch.autograd.Function):
             grad_dist.contiguous(), lam_ext.contiguous(), mu_hist,
             eng.k_conf_indptr, eng.k_conf_base, eng.k_conf_cols,
             eng.link_rates.contiguous(), eng.bw_comp.contiguous(),
-            eng.k_edges, eng.node_vedge, float(eng.T), eng.fp_iters)
+            eng.k_edges, eng.node_vedge, eng.T_arr.contiguous(),
+            eng.fp_iters)
         return dlam, None

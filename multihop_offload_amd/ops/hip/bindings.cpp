@@ -12,23 +12,25 @@ std::vector<torch::Tensor> walk_eval_hip(
     torch::Tensor dl, torch::Tensor adj_indptr, torch::Tensor adj_idx,
     torch::Tensor adj_link, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
-    torch::Tensor bw, torch::Tensor edges, double T, long H, long fp_iters);
+    torch::Tensor bw, torch::Tensor edges, torch::Tensor T_arr, long H,
+    long fp_iters);
 std::vector<torch::Tensor> critic_hip(
     torch::Tensor route_links, torch::Tensor nhop, torch::Tensor vedge_dst,
     torch::Tensor mask, torch::Tensor rate, torch::Tensor ul,
     torch::Tensor dl, torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
-    double T, long Ee, long iters);
+    torch::Tensor T_arr, long Ee, long iters);
 std::vector<torch::Tensor> actor_head_fwd_hip(
     torch::Tensor lam_ext, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
     torch::Tensor bw_comp, torch::Tensor edges, torch::Tensor node_vedge,
-    double T, long N, long iters);
+    torch::Tensor T_arr, long N, long iters);
 torch::Tensor actor_head_bwd_hip(
     torch::Tensor grad_dist, torch::Tensor lam_ext, torch::Tensor mu_hist,
     torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
-    torch::Tensor edges, torch::Tensor node_vedge, double T, long iters);
+    torch::Tensor edges, torch::Tensor node_vedge, torch::Tensor T_arr,
+    long iters);
 
 std::vector<torch::Tensor> cheb_fwd_hip(
     torch::Tensor x, torch::Tensor W, torch::Tensor bias,

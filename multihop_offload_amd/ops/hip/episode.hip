@@ -91,7 +91,8 @@ __global__ void walk_eval_kernel(
     float* __restrict__ unit_mtx,        // (B,N,N) out (prezeroed)
     bool* __restrict__ written,          // (B,N,N) out (prezeroed)
     int* __restrict__ overflow,          // (B) out
-    float T, int N, int E, int J, int H, int fp_iters) {
+    const float* __restrict__ T_arr,     // (B)
+    int N, int E, int J, int H, int fp_iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);          // E
     float* mu = lam + E;                                      // E
@@ -108,6 +109,7 @@ __global__ void walk_eval_kernel(
     const float* ratesb = rates + (size_t)b * E;
     const int* cip = conf_indptr + (size_t)b * (E + 1);
     const int* ccols = conf_cols + conf_base[b];
+    const float T = T_arr[b];
 
     for (int e = tid; e < E; e += nt) lam[e] = 0.0f;
     for (int n = tid; n < N; n += nt) sload[n] = 0.0f;
@@ -224,7 +226,8 @@ std::vector<torch::Tensor> walk_eval_hip(
     torch::Tensor dl, torch::Tensor adj_indptr, torch::Tensor adj_idx,
     torch::Tensor adj_link, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
-    torch::Tensor bw, torch::Tensor edges, double T, long H, long fp_iters) {
+    torch::Tensor bw, torch::Tensor edges, torch::Tensor T_arr, long H,
+    long fp_iters) {
     const int B = sp.size(0), N = sp.size(1);
     const int J = src.size(1), E = rates.size(1);
     auto opts_i = adj_indptr.options();
@@ -255,6 +258,7 @@ std::vector<torch::Tensor> walk_eval_hip(
                        delay_emp.data_ptr<float>(),
                        unit_mtx.data_ptr<float>(), written.data_ptr<bool>(),
                        overflow.data_ptr<int>(),
-                       (float)T, N, E, J, (int)H, (int)fp_iters);
+                       T_arr.data_ptr<float>(), N, E, J, (int)H,
+                       (int)fp_iters);
     return {route_links, nhop, delay_emp, unit_mtx, written, overflow};
 }

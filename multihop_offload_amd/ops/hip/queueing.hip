@@ -119,7 +119,8 @@ __global__ void critic_kernel(
     const float* __restrict__ bw_comp,     // (B,C)
     float* __restrict__ grad_edge,         // (B,Ee) out (prezeroed)
     float* __restrict__ loss_out,          // (B,) out
-    float T, int E, int C, int Ee, int J, int H, int iters) {
+    const float* __restrict__ T_arr,       // (B)
+    int E, int C, int Ee, int J, int H, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam_e = reinterpret_cast<float*>(smem_raw);
     float* unit = lam_e + Ee;
@@ -138,6 +139,7 @@ __global__ void critic_kernel(
     const int* ccols = conf_cols + conf_base[b];
     const float* ratesb = rates + (size_t)b * E;
     const float* bwb = bw_comp + (size_t)b * C;
+    const float T = T_arr[b];
 
     for (int e = tid; e < Ee; e += nt) {
         lam_e[e] = 0.f; dunit[e] = 0.f; dlam[e] = 0.f; dgre[e] = 0.f;
@@ -248,7 +250,8 @@ __global__ void actor_head_fwd_kernel(
     const long* __restrict__ node_vedge,   // (B,N)
     float* __restrict__ dm,                // (B,N,N) out (prezeroed)
     float* __restrict__ mu_hist_out,       // (B,(iters+1),E) out
-    float T, int N, int E, int C, int Ee, int iters) {
+    const float* __restrict__ T_arr, int N, int E, int C, int Ee,
+    int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);       // E
     float* hist = lam + E;
@@ -264,6 +267,7 @@ __global__ void actor_head_fwd_kernel(
     const int* edg = edges + (size_t)b * E * 2;
     const long* nv = node_vedge + (size_t)b * N;
     float* dmb = dm + (size_t)b * N * N;
+    const float T = T_arr[b];
 
     for (int e = tid; e < E; e += nt) lam[e] = le[e];
     __syncthreads();
@@ -298,7 +302,8 @@ __global__ void actor_head_bwd_kernel(
     const int* __restrict__ edges,
     const long* __restrict__ node_vedge,
     float* __restrict__ dlam_ext,          // (B,Ee) out
-    float T, int N, int E, int C, int Ee, int iters) {
+    const float* __restrict__ T_arr, int N, int E, int C, int Ee,
+    int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);
     float* hist = lam + E;
@@ -318,6 +323,7 @@ __global__ void actor_head_bwd_kernel(
     const int* edg = edges + (size_t)b * E * 2;
     const long* nv = node_vedge + (size_t)b * N;
     float* out = dlam_ext + (size_t)b * Ee;
+    const float T = T_arr[b];
 
     for (int e = tid; e < E; e += nt) lam[e] = le[e];
     for (size_t i = tid; i < (size_t)(iters + 1) * E; i += nt)
@@ -362,7 +368,7 @@ std::vector<torch::Tensor> critic_hip(
     torch::Tensor mask, torch::Tensor rate, torch::Tensor ul,
     torch::Tensor dl, torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
-    double T, long Ee, long iters) {
+    torch::Tensor T_arr, long Ee, long iters) {
     const int B = route_links.size(0), J = route_links.size(1);
     const int H = route_links.size(2);
     const int E = rates.size(1), C = bw_comp.size(1);
@@ -383,7 +389,8 @@ std::vector<torch::Tensor> critic_hip(
                        conf_cols.data_ptr<int>(), rates.data_ptr<float>(),
                        bw_comp.data_ptr<float>(),
                        grad_edge.data_ptr<float>(), loss.data_ptr<float>(),
-                       (float)T, E, C, (int)Ee, J, H, (int)iters);
+                       T_arr.data_ptr<float>(), E, C, (int)Ee, J, H,
+                       (int)iters);
     return {grad_edge, loss};
 }
 
@@ -391,7 +398,7 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
     torch::Tensor lam_ext, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
     torch::Tensor bw_comp, torch::Tensor edges, torch::Tensor node_vedge,
-    double T, long N, long iters) {
+    torch::Tensor T_arr, long N, long iters) {
     const int B = lam_ext.size(0), Ee = lam_ext.size(1);
     const int E = rates.size(1), C = bw_comp.size(1);
     auto dm = torch::zeros({B, N, N}, lam_ext.options());
@@ -408,7 +415,8 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
                        bw_comp.data_ptr<float>(), edges.data_ptr<int>(),
                        node_vedge.data_ptr<long>(), dm.data_ptr<float>(),
                        mu_hist.data_ptr<float>(),
-                       (float)T, (int)N, E, C, Ee, (int)iters);
+                       T_arr.data_ptr<float>(), (int)N, E, C, Ee,
+                       (int)iters);
     return {dm, mu_hist};
 }
 
@@ -416,7 +424,8 @@ torch::Tensor actor_head_bwd_hip(
     torch::Tensor grad_dist, torch::Tensor lam_ext, torch::Tensor mu_hist,
     torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
-    torch::Tensor edges, torch::Tensor node_vedge, double T, long iters) {
+    torch::Tensor edges, torch::Tensor node_vedge, torch::Tensor T_arr,
+    long iters) {
     const int B = lam_ext.size(0), Ee = lam_ext.size(1);
     const int E = rates.size(1), C = bw_comp.size(1);
     const int N = grad_dist.size(1);
@@ -433,6 +442,6 @@ torch::Tensor actor_head_bwd_hip(
                        conf_cols.data_ptr<int>(), rates.data_ptr<float>(),
                        bw_comp.data_ptr<float>(), edges.data_ptr<int>(),
                        node_vedge.data_ptr<long>(), dlam.data_ptr<float>(),
-                       (float)T, N, E, C, Ee, (int)iters);
+                       T_arr.data_ptr<float>(), N, E, C, Ee, (int)iters);
     return dlam;
 }

@@ -63,9 +63,11 @@ def fixed_point_mu(lam: torch.Tensor, rates: torch.Tensor, cf_degs: torch.Tensor
     return mu
 
 
-def delay_with_fallback(lam: torch.Tensor, mu: torch.Tensor, T: float,
+def delay_with_fallback(lam: torch.Tensor, mu: torch.Tensor, T,
                         denom: float) -> torch.Tensor:
     """delay = 1/(mu-lam), replaced by T*lam/(denom*mu) where lam > mu.
+    ``T`` may be a scalar or a tensor broadcastable against ``lam``
+    (per-graph horizons).
 
     Matches the TF semantics: the congested entries are *overwritten* by the
     fallback (tensor_scatter_nd_update), so no gradient flows through the
@@ -73,7 +75,7 @@ def delay_with_fallback(lam: torch.Tensor, mu: torch.Tensor, T: float,
     congested = (lam - mu) > 0
     safe = torch.where(congested, torch.ones_like(mu), mu - lam)
     normal = 1.0 / safe
-    fallback = float(T) * lam / (denom * mu)
+    fallback = T * lam / (denom * mu)
     return torch.where(congested, fallback, normal)
 
 
