@@ -217,6 +217,13 @@ class EpisodeEngine:
         import os
         self.use_hip = (self.device.type == "cuda"
                         and os.environ.get("MHO_FORCE_TORCH") != "1")
+        # per-kernel LDS fit (large graphs fall back to the torch tensor
+        # path for that stage; the rest stay on the fused kernels)
+        LDS = 160 * 1024
+        it = fp_iters
+        self.hip_walk_ok = (3 * E + N) * 4 <= LDS
+        self.hip_critic_ok = (5 * Ee + (it + 1) * E + 3 * E) * 4 <= LDS
+        self.hip_actor_ok = ((it + 2) * E + 5 * E) * 4 <= LDS
         if self.use_hip:
             assert dtype == torch.float32, "HIP kernels are fp32"
             from .ops import dispatch
@@ -272,9 +279,11 @@ class EpisodeEngine:
         x = torch.stack([self.f_self_loop, self.f_rate, f_job,
                          self.f_as_server], dim=-1)               # (B,Ē,4)
 
-        if self.use_hip:
+        if self.use_hip and self.hip_actor_ok:
             from .ops.functions import ActorHeadFn, ChebStackFn, cheb_compat
-            if cheb_compat(self.model):
+            rows_pad = (Ee + 15) & ~15
+            cheb_fits = (3 * rows_pad * 33 + 2 * 32 * 32 + 32) * 4 <= 160 * 1024
+            if cheb_compat(self.model) and cheb_fits:
                 params = []
                 for layer in self.model.layers:
                     params += [layer.weight, layer.bias]
@@ -449,7 +458,7 @@ class EpisodeEngine:
                       sp: torch.Tensor):
         """Walk + evaluate: fused HIP kernel on GPU, torch path on CPU.
         Returns (route_links, nhop, delay_emp, unit_mtx, written)."""
-        if self.use_hip:
+        if self.use_hip and self.hip_walk_ok:
             from .ops import dispatch
             ext = dispatch.require_hip()
             H = min(self.walk_cap, 64)
@@ -558,7 +567,7 @@ class EpisodeEngine:
         (grad_edge (B,Ē), loss_fn scalar)."""
         B, E, Ee, J = self.B, self.E, self.Ee, self.Jmax
         vedge_dst = self.node_vedge.gather(1, dst)            # (B,J)
-        if self.use_hip:
+        if self.use_hip and self.hip_critic_ok:
             from .ops import dispatch
             ext = dispatch.require_hip()
             grad_edge, loss = ext.critic(
