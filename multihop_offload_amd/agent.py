@@ -130,6 +130,25 @@ class ACOAgent:
                           g.num_nodes)
         return dm, link_delay, node_delay
 
+    def forward_gcn(self, g: CaseGraph, jobs: JobInstance):
+        """3-feature variant (reference ``forward_gcn``,
+        gnn_offloading_agent.py:171-209): the GNN output is used directly as
+        per-link/per-node delays (no queueing fixed point).  Unused by the
+        harnesses; kept for API parity."""
+        ext = g.ext
+        support, _ = self._support(g)
+        x4 = ext.features(jobs)
+        x = self._t(x4[:, :3])                  # [loop, rate, job]
+        if self.model.layers[0].weight.shape[1] != 3:
+            raise ValueError("forward_gcn needs a model with in_dim=3")
+        lam = self.model(x, support)[:, 0]
+        E = g.num_links
+        dm = delay_matrix(lam[:E], lam[E:],
+                          torch.as_tensor(g.edges, device=self.device),
+                          torch.as_tensor(ext.comp_nodes, device=self.device),
+                          g.num_nodes)
+        return dm
+
     # -- forward + environment (reference :278-291) -----------------------------
     def _env_step(self, env: AdhocCloudEnv, explore: float,
                   rng: Optional[np.random.RandomState] = None):
