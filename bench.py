@@ -48,6 +48,13 @@ def main():
                     help="episodes per GPU per step")
     ap.add_argument("--nodes", type=int, default=100)
     ap.add_argument("--distinct", type=int, default=16)
+    ap.add_argument("--gtype", type=str, default="ba",
+                    help="graph family (BASELINE config 4: --gtype er "
+                         "--nodes 1000 --distinct 1)")
+    ap.add_argument("--sizes", type=str, default=None,
+                    help="BASELINE config 5: comma-separated node counts; "
+                         "one engine per size, per-graph T drawn from "
+                         "{700,800,950,1000}, a step runs all engines")
     ap.add_argument("--T", type=int, default=1000)
     ap.add_argument("--load", type=float, default=0.15)
     ap.add_argument("--K", type=int, default=2)
@@ -67,10 +74,25 @@ def main():
     import torch.distributed as dist
     distributed = dist.is_available() and dist.is_initialized()
 
-    cases = build_cases(args.nodes, args.batch, args.distinct, args.T,
-                        args.seed + rank)          # rank-offset topologies
     model = ChebConvStack(K=args.K, dtype=dtype, seed=args.seed)
-    engine = EpisodeEngine(cases, model, device=device, dtype=dtype)
+    if args.sizes:
+        sizes = [int(s) for s in args.sizes.split(",")]
+        t_choices = [700, 800, 950, 1000]
+        engines = []
+        for i, n in enumerate(sizes):
+            cases = build_cases(n, max(args.batch // len(sizes), 8),
+                                min(args.distinct, 8),
+                                t_choices[i % len(t_choices)],
+                                args.seed + rank + 31 * n, args.gtype)
+            engines.append(EpisodeEngine(cases, model, device=device,
+                                         dtype=dtype))
+        episodes_per_step = sum(e.B for e in engines)
+    else:
+        cases = build_cases(args.nodes, args.batch, args.distinct, args.T,
+                            args.seed + rank, args.gtype)
+        engines = [EpisodeEngine(cases, model, device=device, dtype=dtype)]
+        episodes_per_step = args.batch
+    engine = engines[0]
     dp.broadcast_params(engine.model)
     optimizer = torch.optim.Adam(engine.model.parameters(), lr=1e-4, eps=1e-7)
     reducer = dp.FlatAllreduce(engine.model.parameters())
@@ -78,24 +100,26 @@ def main():
     gen.manual_seed(args.seed * 1009 + rank)
 
     def step():
-        jobs = engine.sample_jobs(args.load, gen)
-        for p in engine.model.parameters():
+        for p in model.parameters():
             p.grad = None
-        res = engine.gnn_episode(jobs, explore=0.0, gen=gen, train=True)
+        res = None
+        for eng in engines:
+            jobs = eng.sample_jobs(args.load, gen)
+            res = eng.gnn_episode(jobs, explore=0.0, gen=gen, train=True)
         # scale summed instance-gradients to a mean; clip like the reference
         with torch.no_grad():
-            for p in engine.model.parameters():
+            for p in model.parameters():
                 if p.grad is not None:
-                    p.grad /= args.batch
+                    p.grad /= episodes_per_step
         reducer(average=True)
         with torch.no_grad():
-            for p in engine.model.parameters():
+            for p in model.parameters():
                 if p.grad is not None:
                     n = p.grad.norm()
                     if n > 1.0:
                         p.grad *= 1.0 / n
         optimizer.step()
-        engine.model.apply_constraints()
+        model.apply_constraints()
         return res
 
     for _ in range(args.warmup):
@@ -121,7 +145,7 @@ def main():
         dist.barrier()
 
     n_gpus = world if distributed else 1
-    episodes = args.batch * n_gpus * args.steps
+    episodes = episodes_per_step * n_gpus * args.steps
     value = episodes / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
     if rank == 0:
@@ -140,8 +164,10 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": f"ChebConv-K{args.K}-5L-h32",
-                "graph": f"ba{args.nodes}-m2",
-                "global_batch": args.batch * n_gpus,
+                "graph": (f"mixed[{args.sizes}]-{args.gtype}" if args.sizes
+                          else f"{args.gtype}{args.nodes}" +
+                          ("-m2" if args.gtype == "ba" else "")),
+                "global_batch": episodes_per_step * n_gpus,
                 "load": args.load,
                 "T": args.T,
                 "parallelism": f"dp{n_gpus}",
