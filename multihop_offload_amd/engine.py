@@ -55,8 +55,8 @@ class EpisodeResult:
     congest: torch.Tensor        # (B,) number of congested jobs
     num_jobs: torch.Tensor       # (B,)
     delay_emp: torch.Tensor      # (B, J) per-job empirical delay (nan-padded)
-    loss_fn: Optional[float] = None
-    loss_mse: Optional[float] = None
+    loss_fn: Optional[torch.Tensor] = None      # 0-dim, on device
+    loss_mse: Optional[torch.Tensor] = None
 
 
 class EpisodeEngine:
@@ -578,7 +578,7 @@ class EpisodeEngine:
                 self.k_conf_cols, self.link_rates.contiguous(),
                 self.bw_comp.contiguous(), self.T_arr.contiguous(), Ee,
                 self.fp_iters)
-            return grad_edge, float(loss.sum())
+            return grad_edge, loss.sum()
         H = route_links.shape[2]
         valid = route_links >= 0
         safe = route_links.clamp(min=0)
@@ -614,7 +614,8 @@ class EpisodeEngine:
         delay_job_edge = torch.maximum(
             jobs_data * unit_edge[..., None] * routes, routes)
         loss_fn = delay_job_edge.sum()
-        (grad_routes,) = torch.autograd.grad(loss_fn, routes)  # (B,Ē,J)
+        (grad_routes,) = torch.autograd.grad(loss_fn, routes,
+                                             retain_graph=False)  # (B,Ē,J)
 
         # route-bias VJP: forward prefix sums of -grad_routes along each route
         # (links in order, then the destination self-loop edge)
@@ -633,7 +634,7 @@ class EpisodeEngine:
         grad_edge = grad_edge.scatter_add(
             0, flat_seq, torch.where(seq_valid, pref,
                                      torch.zeros_like(pref)).reshape(-1))
-        return grad_edge.reshape(B, Ee), float(loss_fn.detach())
+        return grad_edge.reshape(B, Ee), loss_fn.detach()
 
     def grad_dist_matrix(self, grad_edge: torch.Tensor, dm: torch.Tensor,
                          unit_mtx: torch.Tensor, written: torch.Tensor):
@@ -656,9 +657,9 @@ class EpisodeEngine:
         anchor_mask = written & torch.isfinite(dm.detach())
         anchor = torch.where(anchor_mask, 0.001 * diff,
                              torch.zeros_like(diff))
-        n_valid = int(anchor_mask.sum())
-        loss_mse = float((torch.where(anchor_mask, diff, torch.zeros_like(diff))
-                          ** 2).sum() / max(n_valid, 1))
+        n_valid = anchor_mask.sum().clamp(min=1)
+        loss_mse = ((torch.where(anchor_mask, diff, torch.zeros_like(diff))
+                     ** 2).sum() / n_valid.to(diff.dtype))
         return grad_dist + anchor, loss_mse
 
     # ------------------------------------------------------------ episodes

@@ -168,7 +168,7 @@ def main(argv=None):
             njobs = int(res.num_jobs.sum())
             rec = {"step": step, "tau": tau,
                    "congest_ratio": congest / max(njobs, 1),
-                   "loss_fn": res.loss_fn, "loss_mse": res.loss_mse,
+                   "loss_fn": float(res.loss_fn), "loss_mse": float(res.loss_mse),
                    "explore": explore,
                    "eps_per_sec": engine.B * world * step
                    / (time.time() - t0)}

@@ -91,8 +91,8 @@ def test_gnn_episode_matches_oracle(setup):
         emp_oracle > float(g.T))
 
     # losses agree
-    assert np.isclose(res.loss_fn, loss_fn_oracle, rtol=1e-10)
-    assert np.isclose(res.loss_mse, loss_mse_oracle, rtol=1e-8)
+    assert np.isclose(float(res.loss_fn), loss_fn_oracle, rtol=1e-10)
+    assert np.isclose(float(res.loss_mse), loss_mse_oracle, rtol=1e-8)
 
     # gradients agree
     for p, go in zip(engine.model.parameters(), grads_oracle):
@@ -186,3 +186,29 @@ def test_floyd_warshall_vs_scipy():
     got = torch_ref.floyd_warshall(
         torch.tensor(wm, dtype=torch.float64)[None])[0].numpy()
     assert np.allclose(got, want)
+
+
+def test_per_graph_horizon_T():
+    """A batch mixing T=700 and T=1000 must match per-case oracle runs."""
+    from multihop_offload_amd.env import apsp
+    g1, g2 = _case(seed=21), _case(seed=23)
+    g1.T, g2.T = 700, 1000
+    agent = ACOAgent(AgentConfig(T=0, seed=5), 10)   # T comes from cases
+    _wake(agent.model)
+    engine = EpisodeEngine([g1, g2], agent.model, device="cpu",
+                           dtype=torch.float64)
+    j1 = JobInstance.sample(g1.mobile_nodes, 0.6, np.random.RandomState(1))
+    j2 = JobInstance.sample(g2.mobile_nodes, 0.6, np.random.RandomState(2))
+    jb = _jobbatch_from(engine, [j1, j2])
+    res = engine.gnn_episode(jb, train=True)
+
+    for b, (g, j) in enumerate(((g1, j1), (g2, j2))):
+        env = AdhocCloudEnv(g)
+        env.set_jobs(j)
+        a = ACOAgent(AgentConfig(T=g.T, seed=5), 10)
+        _wake(a.model)
+        out = a.forward_backward(env, 0.0, np.random.RandomState(0))
+        emp = delay_empirical(out[1], out[2])
+        got = res.delay_emp[b, :j.num_jobs].numpy()
+        assert np.allclose(got, emp, rtol=1e-10, equal_nan=True)
+        assert res.congest[b].item() == np.count_nonzero(emp > float(g.T))

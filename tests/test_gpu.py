@@ -101,7 +101,7 @@ def test_engine_gpu_matches_cpu():
     res_g = eng_g.gnn_episode(jb_g, train=True)
     tau_c, tau_g = res_c.tau.numpy(), res_g.tau.cpu().numpy()
     assert np.allclose(tau_c, tau_g, rtol=1e-3)
-    assert np.isclose(res_c.loss_fn, res_g.loss_fn, rtol=1e-3)
+    assert np.isclose(float(res_c.loss_fn), float(res_g.loss_fn), rtol=1e-3)
     for pc, pg in zip(model_c.parameters(), model_g.parameters()):
         a, b = pc.grad.numpy(), pg.grad.cpu().numpy()
         denom = max(np.abs(a).max(), 1e-6)
