@@ -60,6 +60,9 @@ def main():
     ap.add_argument("--K", type=int, default=2)
     ap.add_argument("--seed", type=int, default=100)
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--capture", action="store_true",
+                    help="capture the whole training step in a hipGraph "
+                         "and replay it (single-GPU)")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -73,6 +76,7 @@ def main():
     dtype = torch.float32 if device.startswith("cuda") else torch.float64
     import torch.distributed as dist
     distributed = dist.is_available() and dist.is_initialized()
+    capture = args.capture and use_cuda and not distributed
 
     model = ChebConvStack(K=args.K, dtype=dtype, seed=args.seed)
     if args.sizes:
@@ -94,10 +98,16 @@ def main():
         episodes_per_step = args.batch
     engine = engines[0]
     dp.broadcast_params(engine.model)
-    optimizer = torch.optim.Adam(engine.model.parameters(), lr=1e-4, eps=1e-7)
+    optimizer = torch.optim.Adam(engine.model.parameters(), lr=1e-4,
+                                 eps=1e-7, capturable=capture,
+                                 foreach=True)
     reducer = dp.FlatAllreduce(engine.model.parameters())
-    gen = torch.Generator(device=device)
-    gen.manual_seed(args.seed * 1009 + rank)
+    if args.capture and use_cuda:
+        gen = None                     # default generator is graph-safe
+        torch.manual_seed(args.seed * 1009 + rank)
+    else:
+        gen = torch.Generator(device=device)
+        gen.manual_seed(args.seed * 1009 + rank)
 
     def step():
         for p in model.parameters():
@@ -115,15 +125,23 @@ def main():
         with torch.no_grad():
             for p in model.parameters():
                 if p.grad is not None:
-                    n = p.grad.norm()
-                    if n > 1.0:
-                        p.grad *= 1.0 / n
+                    n = p.grad.norm().clamp(min=1e-12)
+                    p.grad *= torch.clamp(n, max=1.0) / n
         optimizer.step()
         model.apply_constraints()
         return res
 
     for _ in range(args.warmup):
         step()
+
+    if capture:
+        # hipGraph capture: the step must be sync-free and use the default
+        # (graph-safe) RNG; replays advance RNG state per torch's graph pool
+        torch.cuda.synchronize()
+        hip_graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(hip_graph):
+            captured = step()
+        step = lambda: hip_graph.replay() or captured  # noqa: E731
 
     if distributed:
         dist.barrier()

@@ -152,9 +152,8 @@ def main(argv=None):
         with torch.no_grad():
             for p in engine.model.parameters():
                 if p.grad is not None:
-                    n = p.grad.norm()
-                    if n > 1.0:
-                        p.grad *= 1.0 / n
+                    n = p.grad.norm().clamp(min=1e-12)
+                    p.grad *= torch.clamp(n, max=1.0) / n
         opt.step()
         engine.model.apply_constraints()
         explore = max(explore * args.explore_decay, 0.001)

@@ -69,9 +69,8 @@ class ChebConvLayer(nn.Module):
         w = self.weight
         norms = torch.sqrt((w * w).sum(dim=0, keepdim=True))
         w.mul_(torch.clamp(norms, max=1.0) / torch.clamp(norms, min=1e-12))
-        bn = self.bias.norm()
-        if bn > 1.0:
-            self.bias.mul_(1.0 / bn)
+        bn = self.bias.norm().clamp(min=1e-12)
+        self.bias.mul_(torch.clamp(bn, max=1.0) / bn)   # branchless, no sync
 
 
 class ChebConvStack(nn.Module):
