@@ -28,13 +28,18 @@ import numpy as np
 import torch
 
 
-def build_cases(n_nodes, batch, distinct, T, seed, gtype="ba", workers=8):
+def build_cases(n_nodes, batch, distinct, T, seed, gtype="ba", workers=None):
     """Synthetic cases of the named config: topologies with the datagen
     role/bandwidth distributions; the batch replicates `distinct`
     topologies with independent link-rate draws (data=synthetic)."""
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
     from multihop_offload_amd.harness.train_batched import \
         build_training_cases
+    if workers is None:
+        # avoid oversubscribing the host CPU when several DP ranks build
+        # their cases concurrently
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        workers = max(2, 16 // max(world, 1))
     return build_training_cases(n_nodes, batch, distinct, T, seed,
                                 gtype=gtype, workers=workers)
 
