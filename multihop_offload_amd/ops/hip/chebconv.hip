@@ -70,12 +70,18 @@ DEV_INLINE void gemm_wgrad(const float* __restrict__ src,
                            int rows_pad, int tid) {
     const int lane = tid & 63;
     const int wid = tid >> 6;
+    const int nw = blockDim.x >> 6;
     const int k_in = lane >> 4;
     const int c_in = lane & 15;
-    for (int t = wid; t < 4; t += (blockDim.x >> 6)) {
-        const int i0 = (t >> 1) * 16, j0 = (t & 1) * 16;
+    // 4 output tiles × 2 row-range halves keeps all 8 waves busy; the two
+    // halves combine through global fp32 atomics (dW is prezeroed)
+    for (int t = wid; t < 8; t += nw) {
+        const int tile = t >> 1, half = t & 1;
+        const int i0 = (tile >> 1) * 16, j0 = (tile & 1) * 16;
+        const int kk0 = half * (rows_pad / 8);
+        const int kk1 = kk0 + rows_pad / 8;
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-        for (int kk = 0; kk < rows_pad / 4; ++kk) {
+        for (int kk = kk0; kk < kk1; ++kk) {
             const int r = kk * 4 + k_in;
             const float a = src[r * STRIDE + i0 + c_in];
             const float b = delta[r * STRIDE + j0 + c_in];
@@ -83,8 +89,8 @@ DEV_INLINE void gemm_wgrad(const float* __restrict__ src,
         }
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-            dw_out[(i0 + (lane >> 4) * 4 + r) * F + j0 + (lane & 15)] =
-                acc[r];
+            atomicAdd(&dw_out[(i0 + (lane >> 4) * 4 + r) * F + j0
+                              + (lane & 15)], acc[r]);
     }
 }
 
@@ -218,8 +224,8 @@ __global__ void cheb_bwd_kernel(
     float* __restrict__ db,              // (B,L,32) out (prezeroed)
     int B, int Ee, int L, int K, int rows_pad) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-    float* Ab = reinterpret_cast<float*>(smem_raw);
-    float* Db = Ab + (size_t)rows_pad * STRIDE;
+    float* Ab = reinterpret_cast<float*>(smem_raw);   // X_l, later dX
+    float* Db = Ab + (size_t)rows_pad * STRIDE;       // current delta
     float* Tb = Db + (size_t)rows_pad * STRIDE;
     float* Wl = Tb + (size_t)rows_pad * STRIDE;   // K*F*F
 
@@ -295,9 +301,11 @@ __global__ void cheb_bwd_kernel(
             spmv(Tb, Ab, ipt, cls, Ee, rows_pad, tid, nt, 1); // dX += A·U
             __syncthreads();
         }
-        // swap: Ab (dX) becomes the delta of the layer below
-        float* tmp = Ab;                          // pointer swap via copy
-        for (int i = tid; i < rows_pad * STRIDE; i += nt) Db[i] = tmp[i];
+        // swap: Ab (dX) becomes the delta of the layer below; the old
+        // delta buffer is reused as the next X_l staging area
+        float* tmp = Ab;
+        Ab = Db;
+        Db = tmp;
         __syncthreads();
     }
 }
