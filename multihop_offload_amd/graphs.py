@@ -208,6 +208,49 @@ class CaseGraph:
         self.link_rates = np.round(np.clip(
             rng.normal(rates, std, size=(self.num_links,)), 0, hi))
 
+    # -- mobility (offloading_v3.py:80-129) -----------------------------------
+    def random_walk(self, ss: float = 0.1, n: int = 10,
+                    rng: Optional[np.random.RandomState] = None):
+        """Perturb n random node positions by N(0, ss) within the bounding
+        box and rebuild the unit-disk connectivity; retries until connected
+        (reference ``random_walk``).  Returns (adj, new_pos)."""
+        if self.pos is None:
+            raise ValueError("random_walk needs node positions")
+        rng = rng or np.random
+        from scipy.sparse import csr_matrix
+        from scipy.sparse.csgraph import connected_components
+        pos = np.asarray(self.pos, dtype=np.float64)
+        b_min, b_max = pos.min() - 0.05, pos.max() + 0.05
+        for _ in range(1000):
+            mask = rng.choice(self.num_nodes, size=n, replace=False)
+            new_pos = pos.copy()
+            new_pos[mask] += rng.normal(0, ss, size=(n, 2))
+            new_pos = new_pos.clip(b_min, b_max)
+            d = np.sqrt(((new_pos[:, None] - new_pos[None, :]) ** 2).sum(-1))
+            adj = (d <= 1.0).astype(np.int8)
+            np.fill_diagonal(adj, 0)
+            ncomp, _ = connected_components(csr_matrix(adj), directed=False)
+            if ncomp == 1:
+                return adj, new_pos
+        raise RuntimeError("random_walk could not find a connected topology")
+
+    def topology_update(self, adj, pos):
+        """Rebuild the topology in place from a new adjacency (reference
+        ``topology_update``).  Returns ``new_links_map``: for each new link,
+        the old link id carrying the same endpoints, or -1."""
+        old_lm = self.link_matrix
+        new = CaseGraph(self.num_nodes, t_max=self.T, seed=self.seed,
+                        m=self.m, gtype=self.gtype, adj=adj, pos=pos,
+                        cf_radius=self.cf_radius)
+        new_links_map = old_lm[new.edges[:, 0], new.edges[:, 1]]
+        # carry roles/bandwidths over; reset per-link state
+        new.roles = self.roles
+        new.proc_bws = self.proc_bws
+        new.servers = self.servers
+        new.relays = self.relays
+        self.__dict__.update(new.__dict__)
+        return new_links_map
+
     def clone_with_rates(self, base_rates,
                          rng: Optional[np.random.RandomState] = None,
                          std: float = 2.0) -> "CaseGraph":

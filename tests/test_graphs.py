@@ -126,3 +126,20 @@ def test_mat_roundtrip(tmp_path):
     assert np.allclose(g.proc_bws, case["nodes_info"][:, 1])
     assert len(g.servers) == int(np.sum(roles == 1))
     assert len(g.relays) == int(np.sum(roles == 2))
+
+
+def test_mobility_random_walk_and_topology_update():
+    rng = np.random.RandomState(0)
+    g = CaseGraph(25, seed=2, gtype="poisson", m=6)
+    g.links_init(50.0, rng=rng)
+    old_edges = {tuple(e) for e in g.edges.tolist()}
+    old_lm = g.link_matrix.copy()
+    adj, pos = g.random_walk(ss=0.05, n=5, rng=rng)
+    m = g.topology_update(adj, pos)
+    assert g.adj.sum() == 2 * g.num_links
+    assert len(m) == g.num_links
+    for l, (u, v) in enumerate(g.edges):
+        if (u, v) in old_edges:
+            assert m[l] == old_lm[u, v]
+        else:
+            assert m[l] == -1
