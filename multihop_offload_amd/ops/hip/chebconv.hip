@@ -94,6 +94,36 @@ DEV_INLINE void gemm_wgrad(const float* __restrict__ src,
     }
 }
 
+// ---- activation tile <-> global copies (coalesced, float4 on global) -----
+// LDS rows have stride 33 (misaligned for vector LDS ops); global rows are
+// dense F=32 floats, 16B-aligned.  Task split (row, 4-col group) puts
+// consecutive lanes on consecutive 16B global segments.
+DEV_INLINE void store_acts(const float* __restrict__ lds,
+                           float* __restrict__ gmem, int Ee, int tid,
+                           int nt) {
+    for (int t = tid; t < Ee * (F / 4); t += nt) {
+        const int r = t >> 3;
+        const int c = (t & 7) * 4;
+        const float* s = lds + r * STRIDE + c;
+        float4 v = {s[0], s[1], s[2], s[3]};
+        *reinterpret_cast<float4*>(gmem + (size_t)r * F + c) = v;
+    }
+}
+
+DEV_INLINE void load_acts(float* __restrict__ lds,
+                          const float* __restrict__ gmem, int Ee,
+                          int rows_pad, int tid, int nt) {
+    for (int t = tid; t < rows_pad * (F / 4); t += nt) {
+        const int r = t >> 3;
+        const int c = (t & 7) * 4;
+        float4 v = {0.f, 0.f, 0.f, 0.f};
+        if (r < Ee)
+            v = *reinterpret_cast<const float4*>(gmem + (size_t)r * F + c);
+        float* d = lds + r * STRIDE + c;
+        d[0] = v.x; d[1] = v.y; d[2] = v.z; d[3] = v.w;
+    }
+}
+
 // ---- SpMV over the support: dst[r][:] (+)= sum_nb src[nb][:] -------------
 // sign=+1: dst = A·src (overwrite); inplace2: dst[r] = 2*(A·src)[r]-dst[r]
 DEV_INLINE void spmv(const float* __restrict__ src, float* __restrict__ dst,
@@ -154,20 +184,18 @@ __global__ void cheb_fwd_kernel(
     const float* xb = x_in + (size_t)b * Ee * 4;
     float* actsb = acts + (size_t)b * (L + 1) * Ee * F;
 
-    // load features (4 real cols, rest zero)
+    // load features (4 real cols as one float4, rest zero)
     for (int r = tid; r < rows_pad; r += nt) {
         float* row = Xb + r * STRIDE;
         for (int c = 0; c < F; ++c) row[c] = 0.f;
         if (r < Ee) {
-            row[0] = xb[r * 4 + 0];
-            row[1] = xb[r * 4 + 1];
-            row[2] = xb[r * 4 + 2];
-            row[3] = xb[r * 4 + 3];
+            const float4 v =
+                *reinterpret_cast<const float4*>(xb + (size_t)r * 4);
+            row[0] = v.x; row[1] = v.y; row[2] = v.z; row[3] = v.w;
         }
     }
     __syncthreads();
-    for (int r = tid; r < Ee; r += nt)
-        for (int c = 0; c < F; ++c) actsb[r * F + c] = Xb[r * STRIDE + c];
+    store_acts(Xb, actsb, Ee, tid, nt);
 
     for (int l = 0; l < L; ++l) {
         // stage weights + bias
@@ -198,10 +226,7 @@ __global__ void cheb_fwd_kernel(
             }
         }
         __syncthreads();
-        for (int r = tid; r < Ee; r += nt)
-            for (int c = 0; c < F; ++c)
-                actsb[((size_t)(l + 1) * Ee + r) * F + c] =
-                    Xb[r * STRIDE + c];
+        store_acts(Xb, actsb + (size_t)(l + 1) * Ee * F, Ee, tid, nt);
         __syncthreads();
     }
     for (int r = tid; r < Ee; r += nt) lam[(size_t)b * Ee + r] =
@@ -247,23 +272,21 @@ __global__ void cheb_bwd_kernel(
 
     for (int l = L - 1; l >= 0; --l) {
         const bool last = (l == L - 1);
-        // activation mask from stored post-act X_{l+1}
-        for (int r = tid; r < Ee; r += nt) {
-            const float* xa = actsb + ((size_t)(l + 1) * Ee + r) * F;
-            float* d = Db + r * STRIDE;
-            for (int c = 0; c < F; ++c)
-                d[c] *= xa[c] > 0.f ? 1.f : (last ? 0.f : 0.2f);
+        // activation mask from stored post-act X_{l+1} (coalesced float4)
+        const float slope = last ? 0.f : 0.2f;
+        for (int t = tid; t < Ee * (F / 4); t += nt) {
+            const int r = t >> 3;
+            const int c = (t & 7) * 4;
+            const float4 v = *reinterpret_cast<const float4*>(
+                actsb + ((size_t)(l + 1) * Ee + r) * F + c);
+            float* d = Db + r * STRIDE + c;
+            d[0] *= v.x > 0.f ? 1.f : slope;
+            d[1] *= v.y > 0.f ? 1.f : slope;
+            d[2] *= v.z > 0.f ? 1.f : slope;
+            d[3] *= v.w > 0.f ? 1.f : slope;
         }
         // load X_l
-        for (int r = tid; r < rows_pad; r += nt) {
-            float* a = Ab + r * STRIDE;
-            if (r < Ee) {
-                const float* xa = actsb + ((size_t)l * Ee + r) * F;
-                for (int c = 0; c < F; ++c) a[c] = xa[c];
-            } else {
-                for (int c = 0; c < F; ++c) a[c] = 0.f;
-            }
-        }
+        load_acts(Ab, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
         for (int i = tid; i < K * F * F; i += nt)
             Wl[i] = W[((size_t)l * K) * F * F + i];
         __syncthreads();

@@ -32,20 +32,42 @@ __global__ void fw_lds_kernel(T* __restrict__ d, int N) {
     const int total = N * N;
     for (int c = tid; c < total; c += nt) s[c] = D[c];
     __syncthreads();
-    // thread layout: column j fixed per thread (nc columns per stripe of
-    // 128 lanes), rows strided — d[k][j] is loaded once per k and d[i][k]
-    // broadcasts across the j-lanes of a wave.
-    const int i0 = tid >> 7;
-    const int istep = nt >> 7;
-    for (int k = 0; k < N; ++k) {
-        for (int j = tid & 127; j < N; j += 128) {
-            const T dkj = s[k * N + j];
-            for (int i = i0; i < N; i += istep) {
-                const T alt = s[i * N + k] + dkj;
-                if (alt < s[i * N + j]) s[i * N + j] = alt;
+    // thread layout: each of 32 lanes per half-wave owns a 4-column group
+    // (vectorized when N % 4 == 0 and fp32), rows strided across the
+    // remaining threads — d[k][j..j+3] loads once per k, d[i][k] broadcasts.
+    const int i0 = tid >> 5;
+    const int istep = nt >> 5;
+    const bool vec4 = (sizeof(T) == 4) && (N % 4 == 0);
+    if (vec4) {
+        for (int k = 0; k < N; ++k) {
+            for (int j = (tid & 31) * 4; j < N; j += 128) {
+                float4 dkj = *reinterpret_cast<const float4*>(
+                    reinterpret_cast<const float*>(s) + k * N + j);
+                for (int i = i0; i < N; i += istep) {
+                    const float dik =
+                        reinterpret_cast<const float*>(s)[i * N + k];
+                    float* row = reinterpret_cast<float*>(s) + i * N + j;
+                    float4 cur = *reinterpret_cast<float4*>(row);
+                    cur.x = fminf(cur.x, dik + dkj.x);
+                    cur.y = fminf(cur.y, dik + dkj.y);
+                    cur.z = fminf(cur.z, dik + dkj.z);
+                    cur.w = fminf(cur.w, dik + dkj.w);
+                    *reinterpret_cast<float4*>(row) = cur;
+                }
             }
+            __syncthreads();
         }
-        __syncthreads();
+    } else {
+        for (int k = 0; k < N; ++k) {
+            for (int j = tid & 127; j < N; j += 128) {
+                const T dkj = s[k * N + j];
+                for (int i = (tid >> 7); i < N; i += (nt >> 7)) {
+                    const T alt = s[i * N + k] + dkj;
+                    if (alt < s[i * N + j]) s[i * N + j] = alt;
+                }
+            }
+            __syncthreads();
+        }
     }
     for (int c = tid; c < total; c += nt) D[c] = s[c];
 }
