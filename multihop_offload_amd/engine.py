@@ -212,6 +212,8 @@ class EpisodeEngine:
             cc.append(np.asarray(ext.ext_indices, dtype=np.int32))
             base.append(base[-1] + len(ext.ext_indices))
         self.k_ext_indptr = t32(ipt)
+        self.k_ext_max_nnz = int(max(int(c.ext.ext_indptr[-1])
+                                     for c in cases))
         self.k_ext_cols = t32(np.concatenate(cc))
         self.k_ext_base = torch.as_tensor(np.asarray(base[:-1]),
                                           dtype=torch.int64,

@@ -35,7 +35,8 @@ class ChebStackFn(torch.autograd.Function):
             Wp[l, :w.shape[0], :w.shape[1], :w.shape[2]] = w
             bp[l, :b.shape[0]] = b
         lam, acts = ext.cheb_fwd(x, Wp, bp, eng.k_ext_indptr,
-                                 eng.k_ext_base, eng.k_ext_cols)
+                                 eng.k_ext_base, eng.k_ext_cols,
+                                 eng.k_ext_max_nnz)
         ctx.save_for_backward(acts, Wp)
         ctx.eng = eng
         ctx.shapes = [tuple(p.shape) for p in params]
@@ -48,7 +49,7 @@ class ChebStackFn(torch.autograd.Function):
         ext = dispatch.require_hip()
         dW, db = ext.cheb_bwd(dlam.contiguous(), acts, Wp,
                               eng.k_ext_indptr, eng.k_ext_base,
-                              eng.k_ext_cols)
+                              eng.k_ext_cols, eng.k_ext_max_nnz)
         dW = dW.sum(dim=0)            # (L,K,32,32) summed over graphs
         db = db.sum(dim=0)            # (L,32)
         grads = []

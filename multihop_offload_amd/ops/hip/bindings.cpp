@@ -34,10 +34,12 @@ torch::Tensor actor_head_bwd_hip(
 
 std::vector<torch::Tensor> cheb_fwd_hip(
     torch::Tensor x, torch::Tensor W, torch::Tensor bias,
-    torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols);
+    torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols,
+    long max_nnz);
 std::vector<torch::Tensor> cheb_bwd_hip(
     torch::Tensor dlam, torch::Tensor acts, torch::Tensor W,
-    torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols);
+    torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols,
+    long max_nnz);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("floyd_warshall", &floyd_warshall_hip);

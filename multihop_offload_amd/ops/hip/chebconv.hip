@@ -169,7 +169,8 @@ __global__ void cheb_fwd_kernel(
     const int* __restrict__ ext_cols,    // flat local
     float* __restrict__ acts,            // (B,L+1,Ee,32) out
     float* __restrict__ lam,             // (B,Ee) out
-    int B, int Ee, int L, int K, int rows_pad) {
+    int B, int Ee, int L, int K, int rows_pad, int max_nnz,
+    int stage_csr) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* Xb = reinterpret_cast<float*>(smem_raw);
     float* Tb = Xb + (size_t)rows_pad * STRIDE;
@@ -181,6 +182,17 @@ __global__ void cheb_fwd_kernel(
     const int tid = threadIdx.x, nt = blockDim.x;
     const int* ipt = ext_indptr + (size_t)b * (Ee + 1);
     const int* cls = ext_cols + ext_base[b];
+    if (stage_csr) {
+        // the support CSR is read by every SpMV of every layer — stage it
+        // in LDS once (latency-bound global gathers otherwise dominate)
+        int* l_ipt = reinterpret_cast<int*>(bl + F);
+        int* l_cols = l_ipt + (Ee + 1);
+        for (int i = tid; i < Ee + 1; i += nt) l_ipt[i] = ipt[i];
+        const int nnz = ipt[Ee];
+        for (int i = tid; i < nnz; i += nt) l_cols[i] = cls[i];
+        ipt = l_ipt;
+        cls = l_cols;
+    }
     const float* xb = x_in + (size_t)b * Ee * 4;
     float* actsb = acts + (size_t)b * (L + 1) * Ee * F;
 
@@ -247,7 +259,8 @@ __global__ void cheb_bwd_kernel(
     const int* __restrict__ ext_cols,
     float* __restrict__ dW,              // (B,L,K,32,32) out (prezeroed)
     float* __restrict__ db,              // (B,L,32) out (prezeroed)
-    int B, int Ee, int L, int K, int rows_pad) {
+    int B, int Ee, int L, int K, int rows_pad, int max_nnz,
+    int stage_csr) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* Ab = reinterpret_cast<float*>(smem_raw);   // X_l, later dX
     float* Db = Ab + (size_t)rows_pad * STRIDE;       // current delta
@@ -258,6 +271,15 @@ __global__ void cheb_bwd_kernel(
     const int tid = threadIdx.x, nt = blockDim.x;
     const int* ipt = ext_indptr + (size_t)b * (Ee + 1);
     const int* cls = ext_cols + ext_base[b];
+    if (stage_csr) {
+        int* l_ipt = reinterpret_cast<int*>(Wl + (size_t)K * F * F);
+        int* l_cols = l_ipt + (Ee + 1);
+        for (int i = tid; i < Ee + 1; i += nt) l_ipt[i] = ipt[i];
+        const int nnz = ipt[Ee];
+        for (int i = tid; i < nnz; i += nt) l_cols[i] = cls[i];
+        ipt = l_ipt;
+        cls = l_cols;
+    }
     const float* actsb = acts + (size_t)b * (L + 1) * Ee * F;
     float* dWb = dW + (size_t)b * L * K * F * F;
     float* dbb = db + (size_t)b * L * F;
@@ -340,16 +362,19 @@ static int round16(int x) { return (x + 15) & ~15; }
 std::vector<torch::Tensor> cheb_fwd_hip(
     torch::Tensor x, torch::Tensor W, torch::Tensor bias,
     torch::Tensor ext_indptr, torch::Tensor ext_base,
-    torch::Tensor ext_cols) {
+    torch::Tensor ext_cols, long max_nnz) {
     const int B = x.size(0), Ee = x.size(1);
     const int L = W.size(0), K = W.size(1);
     TORCH_CHECK(K <= 2, "fused ChebConv kernel supports K<=2");
     const int rows_pad = round16(Ee);
     auto acts = torch::empty({B, L + 1, Ee, F}, x.options());
     auto lam = torch::empty({B, Ee}, x.options());
-    const size_t lds = sizeof(float) *
+    size_t lds = sizeof(float) *
         (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F + F);
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for fused ChebConv");
+    const size_t csr_bytes = sizeof(int) * ((size_t)Ee + 1 + max_nnz);
+    const int stage_csr = (lds + csr_bytes <= 160 * 1024) ? 1 : 0;
+    if (stage_csr) lds += csr_bytes;
     auto stream = at::cuda::getCurrentCUDAStream();
     hipLaunchKernelGGL(cheb_fwd_kernel, dim3(B), dim3(512), lds,
                        stream.stream(),
@@ -357,22 +382,25 @@ std::vector<torch::Tensor> cheb_fwd_hip(
                        bias.data_ptr<float>(), ext_indptr.data_ptr<int>(),
                        ext_base.data_ptr<long>(), ext_cols.data_ptr<int>(),
                        acts.data_ptr<float>(), lam.data_ptr<float>(),
-                       B, Ee, L, K, rows_pad);
+                       B, Ee, L, K, rows_pad, (int)max_nnz, stage_csr);
     return {lam, acts};
 }
 
 std::vector<torch::Tensor> cheb_bwd_hip(
     torch::Tensor dlam, torch::Tensor acts, torch::Tensor W,
     torch::Tensor ext_indptr, torch::Tensor ext_base,
-    torch::Tensor ext_cols) {
+    torch::Tensor ext_cols, long max_nnz) {
     const int B = dlam.size(0), Ee = dlam.size(1);
     const int L = W.size(0), K = W.size(1);
     const int rows_pad = round16(Ee);
     auto dW = torch::zeros({B, L, K, F, F}, dlam.options());
     auto db = torch::zeros({B, L, F}, dlam.options());
-    const size_t lds = sizeof(float) *
+    size_t lds = sizeof(float) *
         (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F);
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for fused ChebConv bwd");
+    const size_t csr_bytes = sizeof(int) * ((size_t)Ee + 1 + max_nnz);
+    const int stage_csr = (lds + csr_bytes <= 160 * 1024) ? 1 : 0;
+    if (stage_csr) lds += csr_bytes;
     auto stream = at::cuda::getCurrentCUDAStream();
     hipLaunchKernelGGL(cheb_bwd_kernel, dim3(B), dim3(512), lds,
                        stream.stream(),
@@ -380,6 +408,6 @@ std::vector<torch::Tensor> cheb_bwd_hip(
                        W.data_ptr<float>(), ext_indptr.data_ptr<int>(),
                        ext_base.data_ptr<long>(), ext_cols.data_ptr<int>(),
                        dW.data_ptr<float>(), db.data_ptr<float>(),
-                       B, Ee, L, K, rows_pad);
+                       B, Ee, L, K, rows_pad, (int)max_nnz, stage_csr);
     return {dW, db};
 }
