@@ -83,6 +83,9 @@ def main():
     distributed = dist.is_available() and dist.is_initialized()
     capture = args.capture and use_cuda and not distributed
 
+    from multihop_offload_amd.ops import dispatch as mho_dispatch
+    use_fused = use_cuda and mho_dispatch.hip_available() \
+        and dtype == torch.float32
     model = ChebConvStack(K=args.K, dtype=dtype, seed=args.seed)
     if args.sizes:
         sizes = [int(s) for s in args.sizes.split(",")]
@@ -102,11 +105,16 @@ def main():
         engines = [EpisodeEngine(cases, model, device=device, dtype=dtype)]
         episodes_per_step = args.batch
     engine = engines[0]
+    if use_fused:
+        from multihop_offload_amd.ops.functions import FusedAdam
+        optimizer = FusedAdam(engine.model, lr=1e-4)
+        reducer = None
+    else:
+        optimizer = torch.optim.Adam(engine.model.parameters(), lr=1e-4,
+                                     eps=1e-7, capturable=capture,
+                                     foreach=True)
+        reducer = dp.FlatAllreduce(engine.model.parameters())
     dp.broadcast_params(engine.model)
-    optimizer = torch.optim.Adam(engine.model.parameters(), lr=1e-4,
-                                 eps=1e-7, capturable=capture,
-                                 foreach=True)
-    reducer = dp.FlatAllreduce(engine.model.parameters())
     if args.capture and use_cuda:
         gen = None                     # default generator is graph-safe
         torch.manual_seed(args.seed * 1009 + rank)
@@ -115,13 +123,23 @@ def main():
         gen.manual_seed(args.seed * 1009 + rank)
 
     def step():
-        for p in model.parameters():
-            p.grad = None
+        if use_fused:
+            optimizer.zero_grad()
+        else:
+            for p in model.parameters():
+                p.grad = None
         res = None
         for eng in engines:
             jobs = eng.sample_jobs(args.load, gen)
             res = eng.gnn_episode(jobs, explore=0.0, gen=gen, train=True)
-        # scale summed instance-gradients to a mean; clip like the reference
+        if use_fused:
+            # one fused kernel: grad mean-scale, per-tensor clipnorm, Adam,
+            # max_norm constraints; flat_g is the single RCCL payload
+            if distributed:
+                dist.all_reduce(optimizer.flat_g)
+            optimizer.step(scale=1.0 / (episodes_per_step * world))
+            return res
+        # torch fallback path (CPU / no HIP)
         with torch.no_grad():
             for p in model.parameters():
                 if p.grad is not None:

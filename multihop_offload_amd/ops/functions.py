@@ -92,3 +92,50 @@ class ActorHeadFn(torch.autograd.Function):
             eng.k_edges, eng.node_vedge, eng.T_arr.contiguous(),
             eng.fp_iters)
         return dlam, None
+
+
+class FusedAdam:
+    """Flat-buffer Adam with the reference's update semantics, fused into one
+    kernel: grad scale → per-tensor clipnorm(1.0) →
+    Adam(eps 1e-7) → Keras max_norm(1.0) constraints.  The model's parameter
+    and gradient tensors become views of two flat buffers; ``flat_g`` is the
+    single RCCL all-reduce payload for data-parallel training."""
+
+    def __init__(self, model, lr=1e-4, betas=(0.9, 0.999), eps=1e-7,
+                 constraints=True):
+        import torch as _t
+        params = list(model.parameters())
+        n = sum(p.numel() for p in params)
+        dev = params[0].device
+        assert params[0].dtype == _t.float32, "FusedAdam is fp32"
+        self.flat_p = _t.zeros(n, dtype=_t.float32, device=dev)
+        self.flat_g = _t.zeros(n, dtype=_t.float32, device=dev)
+        self.m = _t.zeros_like(self.flat_p)
+        self.v = _t.zeros_like(self.flat_p)
+        self.step_dev = _t.zeros(1, dtype=_t.int32, device=dev)
+        segs = []
+        off = 0
+        with _t.no_grad():
+            for p in params:
+                k = p.numel()
+                self.flat_p[off:off + k] = p.detach().reshape(-1)
+                p.data = self.flat_p[off:off + k].view(p.shape)
+                p.grad = self.flat_g[off:off + k].view(p.shape)
+                if p.dim() == 3:
+                    K, fi, fo = p.shape
+                    segs.append([off, k, K, fi, fo])
+                else:
+                    segs.append([off, k, 0, 0, 0])
+                off += k
+        self.seg = _t.tensor(segs, dtype=_t.int32, device=dev)
+        self.lr, self.betas, self.eps = lr, betas, eps
+        self.constraints = constraints
+
+    def zero_grad(self):
+        self.flat_g.zero_()
+
+    def step(self, scale: float = 1.0):
+        ext = dispatch.require_hip()
+        ext.fused_adam(self.flat_p, self.flat_g, self.m, self.v, self.seg,
+                       self.step_dev, scale, self.lr, self.betas[0],
+                       self.betas[1], self.eps, self.constraints)

@@ -41,6 +41,11 @@ std::vector<torch::Tensor> cheb_bwd_hip(
     torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols,
     long max_nnz);
 
+void fused_adam_hip(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                    torch::Tensor v, torch::Tensor seg, torch::Tensor step,
+                    double scale, double lr, double beta1, double beta2,
+                    double eps, bool constraints);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("floyd_warshall", &floyd_warshall_hip);
     m.def("decide", &decide_hip);
@@ -50,4 +55,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("actor_head_bwd", &actor_head_bwd_hip);
     m.def("cheb_fwd", &cheb_fwd_hip);
     m.def("cheb_bwd", &cheb_bwd_hip);
+    m.def("fused_adam", &fused_adam_hip);
 }

@@ -22,6 +22,7 @@ setup(
                 "multihop_offload_amd/ops/hip/episode.hip",
                 "multihop_offload_amd/ops/hip/queueing.hip",
                 "multihop_offload_amd/ops/hip/chebconv.hip",
+                "multihop_offload_amd/ops/hip/optimizer.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3"],
