@@ -34,12 +34,19 @@ def init_from_env() -> tuple[int, int]:
 
 
 def broadcast_params(model: torch.nn.Module, src: int = 0):
-    """Initial weight broadcast so all ranks start identical."""
+    """Initial weight broadcast so all ranks start identical.  Copies back
+    IN PLACE (``vector_to_parameters`` would rebind ``p.data`` and break
+    flat-buffer views, e.g. the fused optimizer's)."""
     if not (dist.is_available() and dist.is_initialized()):
         return
     flat = torch.nn.utils.parameters_to_vector(model.parameters())
     dist.broadcast(flat, src=src)
-    torch.nn.utils.vector_to_parameters(flat, model.parameters())
+    off = 0
+    with torch.no_grad():
+        for p in model.parameters():
+            n = p.numel()
+            p.copy_(flat[off:off + n].view_as(p))
+            off += n
 
 
 class FlatAllreduce:
