@@ -65,22 +65,25 @@ class EpisodeEngine:
                  fp_iters: int = 10, walk_cap: Optional[int] = None):
         assert len(cases) > 0
         N = cases[0].num_nodes
-        E = cases[0].num_links
         T = cases[0].T
         assert all(c.num_nodes == N for c in cases), "batch shares N"
-        assert all(c.num_links == E for c in cases), "batch shares E"
+        # link counts are RAGGED (e.g. distinct ER topologies): E = max,
+        # per-graph E_b in E_arr; padded link slots [E_b, E) are isolated
+        # (rate 1, no conflicts, excluded from every edge-indexed scatter).
+        # Extended-edge ids are renumbered so the virtual (self-loop) edges
+        # of every graph start at E: id < E_b keeps, virtual -> E + rank.
+        E = max(c.num_links for c in cases)
         self.cases = list(cases)
         self.B, self.N, self.E, self.T = len(cases), N, E, T
+        self.E_list = [c.num_links for c in cases]
         self.device = torch.device(device)
         self.dtype = dtype
         self.fp_iters = fp_iters
         self.model = model.to(self.device)
         self.walk_cap = walk_cap or N
 
-        # extended edge dimension is ragged across graphs (relay counts
-        # differ) — pad to the max; padded edges are isolated in the support
-        # and excluded from every scatter by the comp-index tables below
-        Ee = max(c.ext.num_edges_ext for c in cases)
+        Cmax = max(len(c.ext.comp_nodes) for c in cases)
+        Ee = E + Cmax
         self.Ee = Ee
         B = self.B
 
@@ -92,17 +95,43 @@ class EpisodeEngine:
             return torch.as_tensor(np.asarray(a), dtype=torch.int64,
                                    device=self.device)
 
-        self.edges = ti([c.edges for c in cases])              # (B,E,2)
-        self.link_rates = t([c.link_rates for c in cases])     # (B,E)
-        self.cf_degs = t([c.cf_degs for c in cases])
+        def pad_rows(arrs, width, fill, dt=np.float64):
+            out = np.full((B, width), fill, dtype=dt)
+            for b, a in enumerate(arrs):
+                out[b, :len(a)] = a
+            return out
+
+        edges_np = np.zeros((B, E, 2), dtype=np.int64)
+        for b, c in enumerate(cases):
+            edges_np[b, :c.num_links] = c.edges
+        self.edges = ti(edges_np)                              # (B,E,2) pad 0
+        self.link_rates = t(pad_rows([c.link_rates for c in cases], E, 1.0))
+        self.cf_degs = t(pad_rows([c.cf_degs for c in cases], E, 0.0))
         self.proc_bws = t([c.proc_bws for c in cases])         # (B,N)
         self.roles = ti([c.roles for c in cases])
         self.adj = torch.as_tensor(
             np.stack([c.adj for c in cases]) != 0, device=self.device)
         self.link_matrix = ti([c.link_matrix for c in cases])  # (B,N,N)
-        self.node_vedge = ti([c.ext.node_vedge for c in cases])  # (B,N), -1 relay
+        # virtual-edge ids under the ragged renumbering: E + rank
+        nv = -np.ones((B, N), dtype=np.int64)
+        for b, c in enumerate(cases):
+            v = c.ext.node_vedge
+            nv[b] = np.where(v >= 0, v - c.num_links + E, -1)
+        self.node_vedge = ti(nv)                               # (B,N)
         self.comp_mask = self.roles < 2                        # (B,N)
         self.sp_hop = t([c.sp_hop for c in cases])             # (B,N,N)
+        # valid-link selection tables (static): flat (B*E) indices of real
+        # links and their (B*N*N) scatter positions
+        lm = np.zeros((B, E), dtype=bool)
+        for b, c in enumerate(cases):
+            lm[b, :c.num_links] = True
+        self.link_mask = torch.as_tensor(lm, device=self.device)
+        sel_b, sel_e = np.nonzero(lm)
+        self._lk_sel = ti(sel_b * E + sel_e)
+        e0v = edges_np[sel_b, sel_e, 0]
+        e1v = edges_np[sel_b, sel_e, 1]
+        self._lk_lin01 = ti(sel_b * (N * N) + e0v * N + e1v)
+        self._lk_lin10 = ti(sel_b * (N * N) + e1v * N + e0v)
 
         # servers, padded
         S = max(len(c.servers) for c in cases)
@@ -119,10 +148,12 @@ class EpisodeEngine:
         self.Jmax = max(mob_counts) - 1 if max(mob_counts) > 1 else 1
         self.mobile_mask = self.roles == 0
 
-        # flat block-diagonal conflict CSR over B·E (original line graph)
+        # flat block-diagonal conflict CSR over B·E (original line graph);
+        # padded link rows are absent (no conflicts)
         rows, cols = [], []
         for b, c in enumerate(cases):
-            r = np.repeat(np.arange(E), np.diff(c.conf_indptr)) + b * E
+            r = np.repeat(np.arange(c.num_links),
+                          np.diff(c.conf_indptr)) + b * E
             rows.append(r)
             cols.append(c.conf_indices + b * E)
         self.conf = ConflictCSR.__new__(ConflictCSR)
@@ -130,34 +161,46 @@ class EpisodeEngine:
         self.conf.col = ti(np.concatenate(cols))
         self.conf.n = B * E
 
-        # flat block-diagonal extended line-graph CSR over B·Ē (GNN support)
+        # flat block-diagonal extended line-graph CSR over B·Ē (GNN
+        # support), ids remapped (virtual edges at E + rank)
+        def remap(ids, Eb):
+            ids = np.asarray(ids)
+            return np.where(ids < Eb, ids, ids - Eb + E)
+
         rows, cols = [], []
         for b, c in enumerate(cases):
             ext = c.ext
-            ne = ext.num_edges_ext            # ≤ Ee; padded tail is isolated
-            r = np.repeat(np.arange(ne), np.diff(ext.ext_indptr)) + b * Ee
+            Eb = c.num_links
+            ne = ext.num_edges_ext
+            r = remap(np.repeat(np.arange(ne), np.diff(ext.ext_indptr)),
+                      Eb) + b * Ee
             rows.append(r)
-            cols.append(ext.ext_indices + b * Ee)
+            cols.append(remap(ext.ext_indices, Eb) + b * Ee)
         self.support = ConflictCSR.__new__(ConflictCSR)
         self.support.row = ti(np.concatenate(rows))
         self.support.col = ti(np.concatenate(cols))
         self.support.n = B * Ee
 
-        # static feature columns (B,Ē), padded (self_loop=1, rate=1)
-        def pad(vals, fill):
-            out = np.full((B, Ee), fill, dtype=np.float64)
-            for b, v in enumerate(vals):
-                out[b, :len(v)] = v
+        # static feature columns (B,Ē) under the renumbering: real links at
+        # [0,Eb), padded link slots isolated, virtual edges at [E, E+Cb)
+        def feat(attr, link_fill, tail_fill):
+            out = np.full((B, Ee), tail_fill, dtype=np.float64)
+            for b, c in enumerate(cases):
+                v = getattr(c.ext, attr)
+                Eb = c.num_links
+                out[b, :Eb] = v[:Eb]
+                out[b, Eb:E] = link_fill
+                out[b, E:E + len(v) - Eb] = v[Eb:]
             return t(out)
 
-        self.f_self_loop = pad([c.ext.edge_self_loop for c in cases], 1.0)
-        self.f_as_server = pad([c.ext.edge_as_server for c in cases], 0.0)
         for c in cases:
             c.ext.refresh_rates()
-        self.f_rate = pad([c.ext.edge_rate_ext for c in cases], 1.0)
+        self.f_self_loop = feat("edge_self_loop", 0.0, 1.0)
+        self.f_as_server = feat("edge_as_server", 0.0, 0.0)
+        self.f_rate = feat("edge_rate_ext", 1.0, 1.0)
         # proc bws of computing nodes in extended-edge order (B, Ē-E),
         # padded with 1.0 (kept finite; padded slots never reach a scatter)
-        self.Cmax = Ee - E
+        self.Cmax = Cmax
         bw = np.ones((B, self.Cmax), dtype=np.float64)
         for b, c in enumerate(cases):
             bw[b, :len(c.ext.comp_nodes)] = c.proc_bws[c.ext.comp_nodes]
@@ -188,9 +231,24 @@ class EpisodeEngine:
                                    device=self.device)
 
         self.k_adj_indptr = t32([c.adj_indptr for c in cases])     # (B,N+1)
-        self.k_adj_idx = t32([c.adj_indices for c in cases])       # (B,2E)
-        self.k_adj_link = t32([c.adj_link_ids for c in cases])     # (B,2E)
-        self.k_conf_indptr = t32([c.conf_indptr for c in cases])   # (B,E+1)
+
+        def pad_i32(arrs, width):
+            out = np.zeros((B, width), dtype=np.int32)
+            for b, a in enumerate(arrs):
+                out[b, :len(a)] = a
+            return t32(out)
+
+        self.k_adj_idx = pad_i32([c.adj_indices for c in cases], 2 * E)
+        self.k_adj_link = pad_i32([c.adj_link_ids for c in cases], 2 * E)
+        # conflict indptr padded by repeating the last value (rows E_b..E
+        # are empty)
+        cip = np.zeros((B, E + 1), dtype=np.int32)
+        for b, c in enumerate(cases):
+            Eb = c.num_links
+            cip[b, :Eb + 1] = c.conf_indptr
+            cip[b, Eb + 1:] = c.conf_indptr[-1]
+        self.k_conf_indptr = t32(cip)
+        self.k_E_arr = t32([c.num_links for c in cases])           # (B,)
         cc, base = [], [0]
         for c in cases:
             cc.append(np.asarray(c.conf_indices, dtype=np.int32))
@@ -199,18 +257,29 @@ class EpisodeEngine:
         self.k_conf_base = torch.as_tensor(np.asarray(base[:-1]),
                                            dtype=torch.int64,
                                            device=self.device)
-        self.k_edges = t32([c.edges for c in cases])               # (B,E,2)
+        self.k_edges = t32(edges_np)                               # (B,E,2)
         self.k_servers = t32(srv)                                  # (B,S)
-        # extended line-graph CSR (ChebConv support), padded rows empty
+        # extended line-graph CSR (ChebConv support) under the
+        # renumbering: rows [E_b, E) empty, virtual rows at E + rank
         ipt = np.zeros((B, Ee + 1), dtype=np.int32)
         cc, base = [], [0]
         for b, c in enumerate(cases):
             ext = c.ext
-            ne = ext.num_edges_ext
-            ipt[b, :ne + 1] = ext.ext_indptr
-            ipt[b, ne + 1:] = ext.ext_indptr[-1]
-            cc.append(np.asarray(ext.ext_indices, dtype=np.int32))
-            base.append(base[-1] + len(ext.ext_indices))
+            Eb = c.num_links
+            src_ipt = ext.ext_indptr
+            counts = np.zeros(Ee, dtype=np.int64)
+            counts[:Eb] = np.diff(src_ipt)[:Eb]
+            nv = ext.num_edges_ext - Eb
+            counts[E:E + nv] = np.diff(src_ipt)[Eb:]
+            ipt[b, 1:] = np.cumsum(counts)
+            # column ids in row order (rows reordered the same way)
+            colsrc = np.asarray(ext.ext_indices)
+            link_cols = colsrc[:src_ipt[Eb]]
+            virt_cols = colsrc[src_ipt[Eb]:]
+            allc = np.concatenate([link_cols, virt_cols])
+            allc = np.where(allc < Eb, allc, allc - Eb + E)
+            cc.append(allc.astype(np.int32))
+            base.append(base[-1] + len(allc))
         self.k_ext_indptr = t32(ipt)
         self.k_ext_max_nnz = int(max(int(c.ext.ext_indptr[-1])
                                      for c in cases))
@@ -315,14 +384,12 @@ class EpisodeEngine:
     def _delay_matrix(self, link_delay, node_delay):
         B, N, E = self.B, self.N, self.E
         flat = torch.zeros(B * N * N, dtype=self.dtype, device=self.device)
-        e0, e1 = self.edges[..., 0], self.edges[..., 1]
-        base = self._bidx[:, None] * (N * N)
-        flat = flat.index_put(((base + e0 * N + e1).reshape(-1),),
-                              link_delay.reshape(-1))
-        flat = flat.index_put(((base + e1 * N + e0).reshape(-1),),
-                              link_delay.reshape(-1))
+        ld_valid = link_delay.reshape(-1)[self._lk_sel]
+        flat = flat.index_put((self._lk_lin01,), ld_valid)
+        flat = flat.index_put((self._lk_lin10,), ld_valid)
         # diagonal: node delays at computing nodes, +inf at relays
-        diag_idx = base + torch.arange(N, device=self.device)[None, :] * (N + 1)
+        diag_idx = (self._bidx[:, None] * (N * N)
+                    + torch.arange(N, device=self.device)[None, :] * (N + 1))
         inf = torch.full((B, N), float("inf"), dtype=self.dtype,
                          device=self.device)
         flat = flat.index_put((diag_idx.reshape(-1),), inf.reshape(-1))
@@ -645,12 +712,9 @@ class EpisodeEngine:
         (gnn_offloading_agent.py:410-416) + 0.001·MSE anchor (:440-444)."""
         B, N, E = self.B, self.N, self.E
         flat = torch.zeros(B * N * N, dtype=self.dtype, device=self.device)
-        e0, e1 = self.edges[..., 0], self.edges[..., 1]
-        base = self._bidx[:, None] * (N * N)
-        flat = flat.index_put(((base + e0 * N + e1).reshape(-1),),
-                              grad_edge[:, :E].reshape(-1))
-        flat = flat.index_put(((base + e1 * N + e0).reshape(-1),),
-                              grad_edge[:, :E].reshape(-1))
+        ge_valid = grad_edge[:, :E].reshape(-1)[self._lk_sel]
+        flat = flat.index_put((self._lk_lin01,), ge_valid)
+        flat = flat.index_put((self._lk_lin10,), ge_valid)
         lin = (self._comp_b * (N * N) + self._comp_n * (N + 1))
         flat = flat.index_put(
             (lin,), grad_edge[self._comp_b, E + self._comp_r])

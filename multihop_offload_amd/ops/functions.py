@@ -75,7 +75,7 @@ class ActorHeadFn(torch.autograd.Function):
             lam_ext.contiguous(), eng.k_conf_indptr, eng.k_conf_base,
             eng.k_conf_cols, eng.link_rates.contiguous(),
             eng.bw_comp.contiguous(), eng.k_edges, eng.node_vedge,
-            eng.T_arr.contiguous(), eng.N, eng.fp_iters)
+            eng.T_arr.contiguous(), eng.k_E_arr, eng.N, eng.fp_iters)
         ctx.save_for_backward(lam_ext, mu_hist)
         ctx.eng = eng
         return dm
@@ -90,7 +90,7 @@ class ActorHeadFn(torch.autograd.Function):
             eng.k_conf_indptr, eng.k_conf_base, eng.k_conf_cols,
             eng.link_rates.contiguous(), eng.bw_comp.contiguous(),
             eng.k_edges, eng.node_vedge, eng.T_arr.contiguous(),
-            eng.fp_iters)
+            eng.k_E_arr, eng.fp_iters)
         return dlam, None
 
 

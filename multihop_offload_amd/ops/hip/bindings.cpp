@@ -24,13 +24,13 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
     torch::Tensor lam_ext, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
     torch::Tensor bw_comp, torch::Tensor edges, torch::Tensor node_vedge,
-    torch::Tensor T_arr, long N, long iters);
+    torch::Tensor T_arr, torch::Tensor E_arr, long N, long iters);
 torch::Tensor actor_head_bwd_hip(
     torch::Tensor grad_dist, torch::Tensor lam_ext, torch::Tensor mu_hist,
     torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
     torch::Tensor edges, torch::Tensor node_vedge, torch::Tensor T_arr,
-    long iters);
+    torch::Tensor E_arr, long iters);
 
 std::vector<torch::Tensor> cheb_fwd_hip(
     torch::Tensor x, torch::Tensor W, torch::Tensor bias,

@@ -250,8 +250,9 @@ __global__ void actor_head_fwd_kernel(
     const long* __restrict__ node_vedge,   // (B,N)
     float* __restrict__ dm,                // (B,N,N) out (prezeroed)
     float* __restrict__ mu_hist_out,       // (B,(iters+1),E) out
-    const float* __restrict__ T_arr, int N, int E, int C, int Ee,
-    int iters) {
+    const float* __restrict__ T_arr,
+    const int* __restrict__ E_arr,         // (B) real link counts (ragged)
+    int N, int E, int C, int Ee, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);       // E
     float* hist = lam + E;
@@ -273,7 +274,8 @@ __global__ void actor_head_fwd_kernel(
     __syncthreads();
     fixed_point_fwd(lam, ratesb, cip, ccols, hist, busy, E, iters, tid, nt);
     const float* mu_last = hist + (size_t)iters * E;
-    for (int e = tid; e < E; e += nt) {
+    const int Eb = E_arr[b];
+    for (int e = tid; e < Eb; e += nt) {
         const float d = unit_fwd(lam[e], mu_last[e], T, 101.0f);
         const int u = edg[e * 2], v = edg[e * 2 + 1];
         dmb[(size_t)u * N + v] = d;
@@ -302,8 +304,9 @@ __global__ void actor_head_bwd_kernel(
     const int* __restrict__ edges,
     const long* __restrict__ node_vedge,
     float* __restrict__ dlam_ext,          // (B,Ee) out
-    const float* __restrict__ T_arr, int N, int E, int C, int Ee,
-    int iters) {
+    const float* __restrict__ T_arr,
+    const int* __restrict__ E_arr,
+    int N, int E, int C, int Ee, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);
     float* hist = lam + E;
@@ -331,10 +334,15 @@ __global__ void actor_head_bwd_kernel(
     __syncthreads();
     const float* mu_last = hist + (size_t)iters * E;
 
-    // link part: cotangent = gd[u,v] + gd[v,u] (dm wrote both)
+    // link part: cotangent = gd[u,v] + gd[v,u] (dm wrote both);
+    // padded link slots [Eb, E) carry zero cotangent
+    const int Eb = E_arr[b];
     for (int e = tid; e < E; e += nt) {
-        const int u = edg[e * 2], v = edg[e * 2 + 1];
-        const float dd = gd[(size_t)u * N + v] + gd[(size_t)v * N + u];
+        float dd = 0.f;
+        if (e < Eb) {
+            const int u = edg[e * 2], v = edg[e * 2 + 1];
+            dd = gd[(size_t)u * N + v] + gd[(size_t)v * N + u];
+        }
         float dl_, dm_;
         unit_bwd(lam[e], mu_last[e], T, 101.0f, dd, &dl_, &dm_);
         dlam[e] = dl_;
@@ -398,7 +406,7 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
     torch::Tensor lam_ext, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
     torch::Tensor bw_comp, torch::Tensor edges, torch::Tensor node_vedge,
-    torch::Tensor T_arr, long N, long iters) {
+    torch::Tensor T_arr, torch::Tensor E_arr, long N, long iters) {
     const int B = lam_ext.size(0), Ee = lam_ext.size(1);
     const int E = rates.size(1), C = bw_comp.size(1);
     auto dm = torch::zeros({B, N, N}, lam_ext.options());
@@ -415,8 +423,8 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
                        bw_comp.data_ptr<float>(), edges.data_ptr<int>(),
                        node_vedge.data_ptr<long>(), dm.data_ptr<float>(),
                        mu_hist.data_ptr<float>(),
-                       T_arr.data_ptr<float>(), (int)N, E, C, Ee,
-                       (int)iters);
+                       T_arr.data_ptr<float>(), E_arr.data_ptr<int>(),
+                       (int)N, E, C, Ee, (int)iters);
     return {dm, mu_hist};
 }
 
@@ -425,7 +433,7 @@ torch::Tensor actor_head_bwd_hip(
     torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
     torch::Tensor edges, torch::Tensor node_vedge, torch::Tensor T_arr,
-    long iters) {
+    torch::Tensor E_arr, long iters) {
     const int B = lam_ext.size(0), Ee = lam_ext.size(1);
     const int E = rates.size(1), C = bw_comp.size(1);
     const int N = grad_dist.size(1);
@@ -442,6 +450,7 @@ torch::Tensor actor_head_bwd_hip(
                        conf_cols.data_ptr<int>(), rates.data_ptr<float>(),
                        bw_comp.data_ptr<float>(), edges.data_ptr<int>(),
                        node_vedge.data_ptr<long>(), dlam.data_ptr<float>(),
-                       T_arr.data_ptr<float>(), N, E, C, Ee, (int)iters);
+                       T_arr.data_ptr<float>(), E_arr.data_ptr<int>(),
+                       N, E, C, Ee, (int)iters);
     return dlam;
 }

@@ -212,3 +212,64 @@ def test_per_graph_horizon_T():
         got = res.delay_emp[b, :j.num_jobs].numpy()
         assert np.allclose(got, emp, rtol=1e-10, equal_nan=True)
         assert res.congest[b].item() == np.count_nonzero(emp > float(g.T))
+
+
+def _er_case(seed, n=24):
+    rng = np.random.RandomState(100 + seed)
+    g = CaseGraph(n, t_max=1000, seed=seed, gtype="er")
+    g.links_init(50.0, rng=rng)
+    g.add_relay(0)
+    for s in (2, 3, 4):
+        g.add_server(s, 300.0)
+    for v in range(5, n):
+        if g.roles[v] == 0:
+            g.set_mobile_bw(v, 10.0)
+    return g
+
+
+def test_ragged_link_counts_match_per_case_oracle():
+    """Two ER topologies with DIFFERENT edge counts in one engine batch must
+    reproduce each case's oracle run exactly (fp64 CPU)."""
+    g1, g2 = _er_case(2), _er_case(5)
+    assert g1.num_links != g2.num_links     # genuinely ragged
+    agent = ACOAgent(AgentConfig(T=1000, seed=5), 10)
+    _wake(agent.model)
+    engine = EpisodeEngine([g1, g2], agent.model, device="cpu",
+                           dtype=torch.float64)
+    j1 = JobInstance.sample(g1.mobile_nodes, 0.3, np.random.RandomState(1))
+    j2 = JobInstance.sample(g2.mobile_nodes, 0.3, np.random.RandomState(2))
+    jb = _jobbatch_from(engine, [j1, j2])
+    for p in engine.model.parameters():
+        p.grad = None
+    res = engine.gnn_episode(jb, train=True)
+
+    grads_sum = None
+    for b, (g, j) in enumerate(((g1, j1), (g2, j2))):
+        env = AdhocCloudEnv(g)
+        env.set_jobs(j)
+        a = ACOAgent(AgentConfig(T=1000, seed=5), 10)
+        _wake(a.model)
+        out = a.forward_backward(env, 0.0, np.random.RandomState(0))
+        emp = delay_empirical(out[1], out[2])
+        got = res.delay_emp[b, :j.num_jobs].numpy()
+        assert np.allclose(got, emp, rtol=1e-10, equal_nan=True)
+        gset = a.memory[-1][0]
+        grads_sum = (gset if grads_sum is None
+                     else [x + y for x, y in zip(grads_sum, gset)])
+    for p, go in zip(engine.model.parameters(), grads_sum):
+        assert np.allclose(p.grad.numpy(), go.numpy(), rtol=1e-8, atol=1e-12)
+
+    # baseline/local agree too
+    rb = engine.baseline_episode(jb)
+    rl = engine.local_episode(jb)
+    for b, (g, j) in enumerate(((g1, j1), (g2, j2))):
+        env = AdhocCloudEnv(g)
+        env.set_jobs(j)
+        _, dlist, dproc = env.dmtx_baseline()
+        sp = apsp(g, dlist)
+        np.fill_diagonal(sp, np.where(dproc > 0, dproc, g.T))
+        env.offloading(sp, g.sp_hop)
+        ldel, sdel, _ = env.run()
+        assert np.isclose(rb.tau[b].item(),
+                          np.nanmean(delay_empirical(ldel, sdel)),
+                          rtol=1e-10)
