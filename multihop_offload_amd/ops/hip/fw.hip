@@ -140,6 +140,60 @@ __global__ void fw_phase2(T* __restrict__ d, int N, int nb, int kb) {
     if (r < N && c < N) D[r * N + c] = cur[ty * TILE + tx];
 }
 
+// float specialization of phase 3: each thread owns 4 cells (float4 along
+// the column axis), the rowt operand broadcasts — 4x fewer LDS
+// instructions in the dominant phase.  Launch dim3(8, 32).
+__global__ void fw_phase3_f32(float* __restrict__ d, int N, int nb, int kb) {
+    __shared__ float rowt[TILE * TILE];
+    __shared__ float colt[TILE * TILE];
+    float* D = d + (size_t)blockIdx.z * N * N;
+    int ib = blockIdx.y, jb = blockIdx.x;
+    if (ib >= kb) ib += 1;
+    if (jb >= kb) jb += 1;
+    if (ib >= nb || jb >= nb) return;
+    const int tx = threadIdx.x;              // 0..7 → 4 columns each
+    const int ty = threadIdx.y;              // 0..31
+    // cooperative tile loads (4 cols per thread)
+    {
+        const int c4 = tx * 4;
+        int r = ib * TILE + ty, c = kb * TILE + c4;
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+            rowt[ty * TILE + c4 + q] =
+                (r < N && c + q < N) ? D[(size_t)r * N + c + q] : INFINITY;
+        r = kb * TILE + ty;
+        c = jb * TILE + c4;
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+            colt[ty * TILE + c4 + q] =
+                (r < N && c + q < N) ? D[(size_t)r * N + c + q] : INFINITY;
+    }
+    __syncthreads();
+    const int r = ib * TILE + ty;
+    const int c0 = jb * TILE + tx * 4;
+    float4 v = {INFINITY, INFINITY, INFINITY, INFINITY};
+    if (r < N) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+            if (c0 + q < N)
+                (&v.x)[q] = D[(size_t)r * N + c0 + q];
+#pragma unroll
+        for (int k = 0; k < TILE; ++k) {
+            const float dik = rowt[ty * TILE + k];
+            const float4 ckj =
+                *reinterpret_cast<const float4*>(colt + k * TILE + tx * 4);
+            v.x = fminf(v.x, dik + ckj.x);
+            v.y = fminf(v.y, dik + ckj.y);
+            v.z = fminf(v.z, dik + ckj.z);
+            v.w = fminf(v.w, dik + ckj.w);
+        }
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+            if (c0 + q < N)
+                D[(size_t)r * N + c0 + q] = (&v.x)[q];
+    }
+}
+
 template <typename T>
 __global__ void fw_phase3(T* __restrict__ d, int N, int nb, int kb) {
     __shared__ T rowt[TILE * TILE];   // tile (ib, kb)
@@ -194,10 +248,18 @@ torch::Tensor floyd_warshall_hip(torch::Tensor w) {
                                        dim3(nb - 1, 2, B), thr, 0,
                                        stream.stream(),
                                        d.data_ptr<scalar_t>(), N, nb, kb);
-                    hipLaunchKernelGGL(fw_phase3<scalar_t>,
-                                       dim3(nb - 1, nb - 1, B), thr, 0,
-                                       stream.stream(),
-                                       d.data_ptr<scalar_t>(), N, nb, kb);
+                    if constexpr (std::is_same_v<scalar_t, float>) {
+                        hipLaunchKernelGGL(fw_phase3_f32,
+                                           dim3(nb - 1, nb - 1, B),
+                                           dim3(8, 32), 0, stream.stream(),
+                                           d.data_ptr<float>(), N, nb, kb);
+                    } else {
+                        hipLaunchKernelGGL(fw_phase3<scalar_t>,
+                                           dim3(nb - 1, nb - 1, B), thr, 0,
+                                           stream.stream(),
+                                           d.data_ptr<scalar_t>(), N, nb,
+                                           kb);
+                    }
                 }
             }
         }
