@@ -144,8 +144,8 @@ __global__ void fw_phase2(T* __restrict__ d, int N, int nb, int kb) {
 // the column axis), the rowt operand broadcasts — 4x fewer LDS
 // instructions in the dominant phase.  Launch dim3(8, 32).
 __global__ void fw_phase3_f32(float* __restrict__ d, int N, int nb, int kb) {
-    __shared__ float rowt[TILE * TILE];
-    __shared__ float colt[TILE * TILE];
+    __shared__ __attribute__((aligned(16))) float rowt[TILE * TILE];
+    __shared__ __attribute__((aligned(16))) float colt[TILE * TILE];
     float* D = d + (size_t)blockIdx.z * N * N;
     int ib = blockIdx.y, jb = blockIdx.x;
     if (ib >= kb) ib += 1;
