@@ -28,7 +28,8 @@ def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
              device, dtype, workers: int = 8):
     """Returns per-method aggregate {tau, congest_jobs, num_jobs} summed /
     averaged over all (size, case, instance)."""
-    agg = {m: {"tau_sum": 0.0, "tau_n": 0, "congest": 0, "jobs": 0}
+    agg = {m: {"tau_sum": 0.0, "tau_n": 0, "congest": 0, "jobs": 0,
+               "ratio_sum": 0.0, "ratio_n": 0}
            for m in ("baseline", "local", "GNN")}
     per_size = {}
     for n in sizes:
@@ -37,7 +38,8 @@ def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
         engine = EpisodeEngine(cases, model, device=device, dtype=dtype)
         gen = torch.Generator(device=device)
         gen.manual_seed(seed + n)
-        ps = {m: {"tau_sum": 0.0, "tau_n": 0, "congest": 0, "jobs": 0}
+        ps = {m: {"tau_sum": 0.0, "tau_n": 0, "congest": 0, "jobs": 0,
+                  "ratio_sum": 0.0, "ratio_n": 0}
               for m in ("baseline", "local", "GNN")}
         for _ in range(instances):
             jobs = engine.sample_jobs(load, gen)
@@ -46,19 +48,27 @@ def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
                 "local": engine.local_episode(jobs),
                 "GNN": engine.gnn_episode(jobs, train=False),
             }
+            bl = results["baseline"].delay_emp
             for m, res in results.items():
+                # per-task latency ratio vs baseline (notebook cell 16)
+                r = res.delay_emp / bl
+                ok = torch.isfinite(r)
                 for d in (agg[m], ps[m]):
                     d["tau_sum"] += float(torch.nansum(res.tau))
                     d["tau_n"] += int(torch.isfinite(res.tau).sum())
                     d["congest"] += int(res.congest.sum())
                     d["jobs"] += int(res.num_jobs.sum())
+                    d["ratio_sum"] += float(r[ok].sum())
+                    d["ratio_n"] += int(ok.sum())
         per_size[n] = {
             m: {"tau": d["tau_sum"] / max(d["tau_n"], 1),
-                "congest_ratio": d["congest"] / max(d["jobs"], 1)}
+                "congest_ratio": d["congest"] / max(d["jobs"], 1),
+                "latency_ratio": d["ratio_sum"] / max(d["ratio_n"], 1)}
             for m, d in ps.items()}
     summary = {
         m: {"tau": d["tau_sum"] / max(d["tau_n"], 1),
             "congest_ratio": d["congest"] / max(d["jobs"], 1),
+            "latency_ratio": d["ratio_sum"] / max(d["ratio_n"], 1),
             "jobs": d["jobs"]}
         for m, d in agg.items()}
     return summary, per_size
