@@ -293,10 +293,11 @@ class EpisodeEngine:
         # per-kernel LDS fit (large graphs fall back to the torch tensor
         # path for that stage; the rest stay on the fused kernels)
         LDS = 160 * 1024
-        it = fp_iters
         self.hip_walk_ok = (3 * E + N) * 4 <= LDS
-        self.hip_critic_ok = (5 * Ee + (it + 1) * E + 3 * E) * 4 <= LDS
-        self.hip_actor_ok = ((it + 2) * E + 5 * E) * 4 <= LDS
+        # the critic/actor kernels switch to global scratch for large
+        # graphs; these are the limits of that mode
+        self.hip_critic_ok = (2 * Ee + 3 * E) * 4 <= LDS
+        self.hip_actor_ok = 5 * E * 4 <= LDS
         if self.use_hip:
             assert dtype == torch.float32, "HIP kernels are fp32"
             from .ops import dispatch

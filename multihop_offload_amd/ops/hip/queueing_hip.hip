@@ -120,15 +120,36 @@ __global__ void critic_kernel(
     float* __restrict__ grad_edge,         // (B,Ee) out (prezeroed)
     float* __restrict__ loss_out,          // (B,) out
     const float* __restrict__ T_arr,       // (B)
+    float* __restrict__ g_hist,            // (B,(iters+1),E) scratch | null
+    float* __restrict__ g_dunit,           // (B,Ee) scratch (prezeroed)
+    float* __restrict__ g_dlam,            // (B,Ee) scratch (prezeroed)
+    int large,                             // 1: use global scratch
     int E, int C, int Ee, int J, int H, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    const int b_ = blockIdx.x;
     float* lam_e = reinterpret_cast<float*>(smem_raw);
     float* unit = lam_e + Ee;
-    float* dunit = unit + Ee;
-    float* dlam = dunit + Ee;
-    float* dgre = dlam + Ee;
-    float* hist = dgre + Ee;
-    float* s1 = hist + (size_t)(iters + 1) * E;
+    float* dunit;
+    float* dlam;
+    float* dgre;
+    float* hist;
+    float* s1;
+    if (large) {
+        // large graphs (e.g. 1000-node ER): reverse-pass accumulators and
+        // the mu history live in global scratch (same-CU coherence through
+        // __syncthreads suffices — one workgroup per graph)
+        dunit = g_dunit + (size_t)b_ * Ee;
+        dlam = g_dlam + (size_t)b_ * Ee;
+        dgre = grad_edge + (size_t)b_ * Ee;   // accumulate output directly
+        hist = g_hist + (size_t)b_ * (iters + 1) * E;
+        s1 = unit + Ee;
+    } else {
+        dunit = unit + Ee;
+        dlam = dunit + Ee;
+        dgre = dlam + Ee;
+        hist = dgre + Ee;
+        s1 = hist + (size_t)(iters + 1) * E;
+    }
     float* s2 = s1 + E;
     float* s3 = s2 + E;
     __shared__ float loss_acc;
@@ -142,7 +163,9 @@ __global__ void critic_kernel(
     const float T = T_arr[b];
 
     for (int e = tid; e < Ee; e += nt) {
-        lam_e[e] = 0.f; dunit[e] = 0.f; dlam[e] = 0.f; dgre[e] = 0.f;
+        lam_e[e] = 0.f;
+        if (!large) { dunit[e] = 0.f; dlam[e] = 0.f; dgre[e] = 0.f; }
+        // (large mode: the global scratch arrives prezeroed)
     }
     if (tid == 0) loss_acc = 0.f;
     __syncthreads();
@@ -230,8 +253,10 @@ __global__ void critic_kernel(
         }
     }
     __syncthreads();
-    float* geb = grad_edge + (size_t)b * Ee;
-    for (int e = tid; e < Ee; e += nt) geb[e] = dgre[e];
+    if (!large) {
+        float* geb = grad_edge + (size_t)b * Ee;
+        for (int e = tid; e < Ee; e += nt) geb[e] = dgre[e];
+    }
     if (tid == 0) loss_out[b] = loss_acc;
 }
 
@@ -252,11 +277,18 @@ __global__ void actor_head_fwd_kernel(
     float* __restrict__ mu_hist_out,       // (B,(iters+1),E) out
     const float* __restrict__ T_arr,
     const int* __restrict__ E_arr,         // (B) real link counts (ragged)
-    int N, int E, int C, int Ee, int iters) {
+    int large, int N, int E, int C, int Ee, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);       // E
-    float* hist = lam + E;
-    float* busy = hist + (size_t)(iters + 1) * E;
+    float* hist;
+    float* busy;
+    if (large) {
+        hist = mu_hist_out + (size_t)blockIdx.x * (iters + 1) * E;
+        busy = lam + E;
+    } else {
+        hist = lam + E;
+        busy = hist + (size_t)(iters + 1) * E;
+    }
 
     const int b = blockIdx.x;
     const int tid = threadIdx.x, nt = blockDim.x;
@@ -286,8 +318,11 @@ __global__ void actor_head_fwd_kernel(
         dmb[(size_t)n * N + n] =
             ve >= 0 ? unit_fwd(le[ve], bwb[ve - E], T, 100.0f) : INFINITY;
     }
-    float* ho = mu_hist_out + (size_t)b * (iters + 1) * E;
-    for (size_t i = tid; i < (size_t)(iters + 1) * E; i += nt) ho[i] = hist[i];
+    if (!large) {
+        float* ho = mu_hist_out + (size_t)b * (iters + 1) * E;
+        for (size_t i = tid; i < (size_t)(iters + 1) * E; i += nt)
+            ho[i] = hist[i];
+    }
 }
 
 // actor head backward: grad_dist (B,N,N) → δλ_ext (B,Ee)
@@ -306,11 +341,19 @@ __global__ void actor_head_bwd_kernel(
     float* __restrict__ dlam_ext,          // (B,Ee) out
     const float* __restrict__ T_arr,
     const int* __restrict__ E_arr,
-    int N, int E, int C, int Ee, int iters) {
+    int large, int N, int E, int C, int Ee, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);
-    float* hist = lam + E;
-    float* dmu = hist + (size_t)(iters + 1) * E;
+    const float* hist;
+    float* dmu;
+    if (large) {
+        hist = mu_hist + (size_t)blockIdx.x * (iters + 1) * E;
+        dmu = lam + E;
+    } else {
+        float* h = lam + E;
+        hist = h;
+        dmu = h + (size_t)(iters + 1) * E;
+    }
     float* dlam = dmu + E;
     float* s1 = dlam + E;
     float* s2 = s1 + E;
@@ -329,8 +372,11 @@ __global__ void actor_head_bwd_kernel(
     const float T = T_arr[b];
 
     for (int e = tid; e < E; e += nt) lam[e] = le[e];
-    for (size_t i = tid; i < (size_t)(iters + 1) * E; i += nt)
-        hist[i] = mu_hist[(size_t)b * (iters + 1) * E + i];
+    if (!large) {
+        float* h = lam + E;
+        for (size_t i = tid; i < (size_t)(iters + 1) * E; i += nt)
+            h[i] = mu_hist[(size_t)b * (iters + 1) * E + i];
+    }
     __syncthreads();
     const float* mu_last = hist + (size_t)iters * E;
 
@@ -382,9 +428,23 @@ std::vector<torch::Tensor> critic_hip(
     const int E = rates.size(1), C = bw_comp.size(1);
     auto grad_edge = torch::zeros({B, (long)Ee}, rates.options());
     auto loss = torch::zeros({B}, rates.options());
-    const size_t lds = sizeof(float) *
+    size_t lds = sizeof(float) *
         (5 * (size_t)Ee + (size_t)(iters + 1) * E + 3 * (size_t)E);
-    TORCH_CHECK(lds <= 160 * 1024, "graph too large for LDS critic kernel");
+    int large = 0;
+    torch::Tensor g_hist, g_dunit, g_dlam;
+    if (lds > 160 * 1024) {
+        large = 1;
+        lds = sizeof(float) * (2 * (size_t)Ee + 3 * (size_t)E);
+        TORCH_CHECK(lds <= 160 * 1024,
+                    "graph too large even for the global-scratch critic");
+        g_hist = torch::empty({B, iters + 1, (long)E}, rates.options());
+        g_dunit = torch::zeros({B, (long)Ee}, rates.options());
+        g_dlam = torch::zeros({B, (long)Ee}, rates.options());
+    } else {
+        g_hist = torch::empty({1}, rates.options());
+        g_dunit = g_hist;
+        g_dlam = g_hist;
+    }
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     hipLaunchKernelGGL(critic_kernel, dim3(B), dim3(256), lds,
                        stream.stream(),
@@ -397,8 +457,10 @@ std::vector<torch::Tensor> critic_hip(
                        conf_cols.data_ptr<int>(), rates.data_ptr<float>(),
                        bw_comp.data_ptr<float>(),
                        grad_edge.data_ptr<float>(), loss.data_ptr<float>(),
-                       T_arr.data_ptr<float>(), E, C, (int)Ee, J, H,
-                       (int)iters);
+                       T_arr.data_ptr<float>(),
+                       g_hist.data_ptr<float>(), g_dunit.data_ptr<float>(),
+                       g_dlam.data_ptr<float>(), large,
+                       E, C, (int)Ee, J, H, (int)iters);
     return {grad_edge, loss};
 }
 
@@ -411,7 +473,12 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
     const int E = rates.size(1), C = bw_comp.size(1);
     auto dm = torch::zeros({B, N, N}, lam_ext.options());
     auto mu_hist = torch::empty({B, iters + 1, (long)E}, lam_ext.options());
-    const size_t lds = sizeof(float) * ((size_t)(iters + 2) * E + E);
+    size_t lds = sizeof(float) * ((size_t)(iters + 2) * E + E);
+    int large = 0;
+    if (lds > 160 * 1024) {
+        large = 1;
+        lds = sizeof(float) * 2 * (size_t)E;
+    }
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for LDS actor head");
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     hipLaunchKernelGGL(actor_head_fwd_kernel, dim3(B), dim3(256), lds,
@@ -424,7 +491,7 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
                        node_vedge.data_ptr<long>(), dm.data_ptr<float>(),
                        mu_hist.data_ptr<float>(),
                        T_arr.data_ptr<float>(), E_arr.data_ptr<int>(),
-                       (int)N, E, C, Ee, (int)iters);
+                       large, (int)N, E, C, Ee, (int)iters);
     return {dm, mu_hist};
 }
 
@@ -438,7 +505,12 @@ torch::Tensor actor_head_bwd_hip(
     const int E = rates.size(1), C = bw_comp.size(1);
     const int N = grad_dist.size(1);
     auto dlam = torch::zeros_like(lam_ext);
-    const size_t lds = sizeof(float) * ((size_t)(iters + 1) * E + 5 * (size_t)E);
+    size_t lds = sizeof(float) * ((size_t)(iters + 1) * E + 5 * (size_t)E);
+    int large = 0;
+    if (lds > 160 * 1024) {
+        large = 1;
+        lds = sizeof(float) * 5 * (size_t)E;
+    }
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for LDS actor head bwd");
     auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
     hipLaunchKernelGGL(actor_head_bwd_kernel, dim3(B), dim3(256), lds,
@@ -451,6 +523,6 @@ torch::Tensor actor_head_bwd_hip(
                        bw_comp.data_ptr<float>(), edges.data_ptr<int>(),
                        node_vedge.data_ptr<long>(), dlam.data_ptr<float>(),
                        T_arr.data_ptr<float>(), E_arr.data_ptr<int>(),
-                       N, E, C, Ee, (int)iters);
+                       large, N, E, C, Ee, (int)iters);
     return dlam;
 }

@@ -251,3 +251,45 @@ def test_ragged_link_counts_gpu_matches_cpu():
     for pc, pg in zip(mc.parameters(), mg.parameters()):
         a, b = pc.grad.numpy(), pg.grad.cpu().numpy()
         assert np.abs(a - b).max() / max(np.abs(a).max(), 1e-6) < 1e-2
+
+
+@needs_gpu
+def test_large_mode_queueing_kernels_match_cpu():
+    """A graph big enough to force the global-scratch critic/actor-head
+    kernel modes must still match the CPU torch path (incl. gradients)."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.graphs import CaseGraph, JobInstance
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from tests.test_engine import _jobbatch_from
+
+    rng = np.random.RandomState(0)
+    g = CaseGraph(400, t_max=1000, seed=11, gtype="er")
+    g.links_init(50.0, rng=rng)
+    g.add_relay(0)
+    for s in range(2, 10):
+        g.add_server(s, 300.0)
+    for v in range(10, 400):
+        if g.roles[v] == 0:
+            g.set_mobile_bw(v, 10.0)
+    cases = [g]
+
+    mc = ChebConvStack(K=2, dtype=torch.float32, seed=1)
+    mg = ChebConvStack(K=2, dtype=torch.float32, seed=1)
+    with torch.no_grad():
+        for pc, pg in zip(mc.parameters(), mg.parameters()):
+            pc.mul_(0.01)
+            pg.copy_(pc)
+        mc.layers[-1].bias.fill_(0.5)
+        mg.layers[-1].bias.fill_(0.5)
+    ec = EpisodeEngine(cases, mc, device="cpu", dtype=torch.float32)
+    eg = EpisodeEngine(cases, mg, device="cuda", dtype=torch.float32)
+    # the point of this test: the GPU engine must be using the kernels
+    assert eg.hip_critic_ok and eg.hip_actor_ok and eg.hip_walk_ok
+    insts = [JobInstance.sample(g.mobile_nodes, 0.2,
+                                np.random.RandomState(3))]
+    rc = ec.gnn_episode(_jobbatch_from(ec, insts), train=True)
+    rg = eg.gnn_episode(_jobbatch_from(eg, insts), train=True)
+    assert np.allclose(rc.tau.numpy(), rg.tau.cpu().numpy(), rtol=2e-3)
+    for pc, pg in zip(mc.parameters(), mg.parameters()):
+        a, b = pc.grad.numpy(), pg.grad.cpu().numpy()
+        assert np.abs(a - b).max() / max(np.abs(a).max(), 1e-6) < 2e-2
