@@ -355,9 +355,7 @@ class EpisodeEngine:
 
         if self.use_hip and self.hip_actor_ok:
             from .ops.functions import ActorHeadFn, ChebStackFn, cheb_compat
-            rows_pad = (Ee + 15) & ~15
-            cheb_fits = (3 * rows_pad * 33 + 2 * 32 * 32 + 32) * 4 <= 160 * 1024
-            if cheb_compat(self.model) and cheb_fits:
+            if cheb_compat(self.model):
                 params = []
                 for layer in self.model.layers:
                     params += [layer.weight, layer.bias]

@@ -19,9 +19,16 @@ def cheb_compat(model) -> bool:
     return True
 
 
+def cheb_lds_fits(Ee: int, K: int = 2) -> bool:
+    rows_pad = (Ee + 15) & ~15
+    return (3 * rows_pad * 33 + K * 32 * 32 + 32) * 4 <= 160 * 1024
+
+
 class ChebStackFn(torch.autograd.Function):
-    """Fused L-layer ChebConv stack (ops/hip/chebconv.hip): x (B,Ē,4) →
-    λ (B,Ē) with per-layer activations saved for the fused backward."""
+    """Fused L-layer ChebConv stack: x (B,Ē,4) → λ (B,Ē) with per-layer
+    activations saved for the fused backward.  Dispatches between the
+    LDS-resident kernel (ops/hip/chebconv.hip) and the row-tiled
+    large-graph kernels (ops/hip/chebconv_large.hip) by LDS fit."""
 
     @staticmethod
     def forward(ctx, x: torch.Tensor, eng, *params):
@@ -34,9 +41,14 @@ class ChebStackFn(torch.autograd.Function):
             w, b = params[2 * l], params[2 * l + 1]
             Wp[l, :w.shape[0], :w.shape[1], :w.shape[2]] = w
             bp[l, :b.shape[0]] = b
-        lam, acts = ext.cheb_fwd(x, Wp, bp, eng.k_ext_indptr,
-                                 eng.k_ext_base, eng.k_ext_cols,
-                                 eng.k_ext_max_nnz)
+        ctx.large = not cheb_lds_fits(eng.Ee, K)
+        if ctx.large:
+            lam, acts = ext.cheb_large_fwd(x, Wp, bp, eng.k_ext_indptr,
+                                           eng.k_ext_base, eng.k_ext_cols)
+        else:
+            lam, acts = ext.cheb_fwd(x, Wp, bp, eng.k_ext_indptr,
+                                     eng.k_ext_base, eng.k_ext_cols,
+                                     eng.k_ext_max_nnz)
         ctx.save_for_backward(acts, Wp)
         ctx.eng = eng
         ctx.shapes = [tuple(p.shape) for p in params]
@@ -47,9 +59,14 @@ class ChebStackFn(torch.autograd.Function):
         acts, Wp = ctx.saved_tensors
         eng = ctx.eng
         ext = dispatch.require_hip()
-        dW, db = ext.cheb_bwd(dlam.contiguous(), acts, Wp,
-                              eng.k_ext_indptr, eng.k_ext_base,
-                              eng.k_ext_cols, eng.k_ext_max_nnz)
+        if ctx.large:
+            dW, db = ext.cheb_large_bwd(dlam.contiguous(), acts, Wp,
+                                        eng.k_ext_indptr, eng.k_ext_base,
+                                        eng.k_ext_cols)
+        else:
+            dW, db = ext.cheb_bwd(dlam.contiguous(), acts, Wp,
+                                  eng.k_ext_indptr, eng.k_ext_base,
+                                  eng.k_ext_cols, eng.k_ext_max_nnz)
         dW = dW.sum(dim=0)            # (L,K,32,32) summed over graphs
         db = db.sum(dim=0)            # (L,32)
         grads = []

@@ -41,6 +41,14 @@ std::vector<torch::Tensor> cheb_bwd_hip(
     torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols,
     long max_nnz);
 
+std::vector<torch::Tensor> cheb_large_fwd_hip(
+    torch::Tensor x, torch::Tensor W, torch::Tensor bias,
+    torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor ext_cols);
+std::vector<torch::Tensor> cheb_large_bwd_hip(
+    torch::Tensor dlam, torch::Tensor acts, torch::Tensor W,
+    torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor ext_cols);
 void fused_adam_hip(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                     torch::Tensor v, torch::Tensor seg, torch::Tensor step,
                     double scale, double lr, double beta1, double beta2,
@@ -56,4 +64,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("cheb_fwd", &cheb_fwd_hip);
     m.def("cheb_bwd", &cheb_bwd_hip);
     m.def("fused_adam", &fused_adam_hip);
+    m.def("cheb_large_fwd", &cheb_large_fwd_hip);
+    m.def("cheb_large_bwd", &cheb_large_bwd_hip);
 }

@@ -23,6 +23,7 @@ setup(
                 "multihop_offload_amd/ops/hip/queueing.hip",
                 "multihop_offload_amd/ops/hip/chebconv.hip",
                 "multihop_offload_amd/ops/hip/optimizer.hip",
+                "multihop_offload_amd/ops/hip/chebconv_large.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3"],
