@@ -293,3 +293,29 @@ def test_large_mode_queueing_kernels_match_cpu():
     for pc, pg in zip(mc.parameters(), mg.parameters()):
         a, b = pc.grad.numpy(), pg.grad.cpu().numpy()
         assert np.abs(a - b).max() / max(np.abs(a).max(), 1e-6) < 2e-2
+
+
+@needs_gpu
+def test_reference_checkpoint_in_engine_gpu():
+    """The reference's shipped trained model (K=1 TF bundle) runs through
+    the fused GPU engine and reproduces its local-collapse behavior."""
+    import os
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.utils.tfckpt import load_reference_weights
+    prefix = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "artifacts", "reference_ckpt",
+        "cp-0000.ckpt")
+    if not os.path.isfile(prefix + ".index"):
+        pytest.skip("reference checkpoint not present")
+    model = ChebConvStack(K=1, dtype=torch.float32)
+    load_reference_weights(model, prefix)
+    cases = _cases(n=30, B=16)
+    engine = EpisodeEngine(cases, model, device="cuda", dtype=torch.float32)
+    gen = torch.Generator(device="cuda")
+    gen.manual_seed(0)
+    jobs = engine.sample_jobs(0.15, gen)
+    res = engine.gnn_episode(jobs, train=False)
+    rl = engine.local_episode(jobs)
+    # their trained policy collapses to local computing on these loads
+    assert torch.allclose(res.tau, rl.tau, rtol=1e-3)
