@@ -67,6 +67,8 @@ def build_training_cases(n_nodes, batch, distinct, T, seed, gtype="ba",
             else:
                 g.set_mobile_bw(nidx, bw)
         g.links_init(case["link_rate"], rng=rng)
+        g.sp_hop        # materialize before cloning so replicas share it
+        g.ext
         bases.append((g, case["link_rate"]))
 
     cases = []
