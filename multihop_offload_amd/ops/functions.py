@@ -46,9 +46,10 @@ class ChebStackFn(torch.autograd.Function):
             lam, acts = ext.cheb_large_fwd(x, Wp, bp, eng.k_ext_indptr,
                                            eng.k_ext_base, eng.k_ext_cols)
         else:
-            lam, acts = ext.cheb_fwd(x, Wp, bp, eng.k_ext_indptr,
-                                     eng.k_ext_base, eng.k_ext_cols,
-                                     eng.k_ext_max_nnz)
+            lam, acts, t1s = ext.cheb_fwd(x, Wp, bp, eng.k_ext_indptr,
+                                          eng.k_ext_base, eng.k_ext_cols,
+                                          eng.k_ext_max_nnz)
+            ctx.t1s = t1s
         ctx.save_for_backward(acts, Wp)
         ctx.eng = eng
         ctx.shapes = [tuple(p.shape) for p in params]
@@ -64,7 +65,7 @@ class ChebStackFn(torch.autograd.Function):
                                         eng.k_ext_indptr, eng.k_ext_base,
                                         eng.k_ext_cols)
         else:
-            dW, db = ext.cheb_bwd(dlam.contiguous(), acts, Wp,
+            dW, db = ext.cheb_bwd(dlam.contiguous(), acts, ctx.t1s, Wp,
                                   eng.k_ext_indptr, eng.k_ext_base,
                                   eng.k_ext_cols, eng.k_ext_max_nnz)
         dW = dW.sum(dim=0)            # (L,K,32,32) summed over graphs

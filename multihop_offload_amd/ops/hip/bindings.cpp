@@ -37,9 +37,9 @@ std::vector<torch::Tensor> cheb_fwd_hip(
     torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols,
     long max_nnz);
 std::vector<torch::Tensor> cheb_bwd_hip(
-    torch::Tensor dlam, torch::Tensor acts, torch::Tensor W,
-    torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols,
-    long max_nnz);
+    torch::Tensor dlam, torch::Tensor acts, torch::Tensor t1s,
+    torch::Tensor W, torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor ext_cols, long max_nnz);
 
 std::vector<torch::Tensor> cheb_large_fwd_hip(
     torch::Tensor x, torch::Tensor W, torch::Tensor bias,

@@ -168,6 +168,7 @@ __global__ void cheb_fwd_kernel(
     const long* __restrict__ ext_base,   // (B)
     const int* __restrict__ ext_cols,    // flat local
     float* __restrict__ acts,            // (B,L+1,Ee,32) out
+    float* __restrict__ t1s,             // (B,L,Ee,32) out: A·X per layer
     float* __restrict__ lam,             // (B,Ee) out
     int B, int Ee, int L, int K, int rows_pad, int max_nnz,
     int stage_csr) {
@@ -223,6 +224,9 @@ __global__ void cheb_fwd_kernel(
             spmv(Xb, Tb, ipt, cls, Ee, rows_pad, tid, nt, 0); // T1 = A·X
             __syncthreads();
             gemm_acc(Tb, Yb, Wl + F * F, false, rows_pad, tid);
+            // save T1 for the backward weight-gradient pass (skips the
+            // SpMV recompute there)
+            store_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee, tid, nt);
             __syncthreads();
         }
         // epilogue: bias + activation → Xb (next layer input)
@@ -253,6 +257,7 @@ __global__ void cheb_fwd_kernel(
 __global__ void cheb_bwd_kernel(
     const float* __restrict__ dlam,      // (B,Ee)
     const float* __restrict__ acts,      // (B,L+1,Ee,32)
+    const float* __restrict__ t1s,       // (B,L,Ee,32): A·X from forward
     const float* __restrict__ W,         // (L,K,32,32)
     const int* __restrict__ ext_indptr,
     const long* __restrict__ ext_base,
@@ -324,7 +329,8 @@ __global__ void cheb_bwd_kernel(
         gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
         if (K > 1) {
             __syncthreads();
-            spmv(Ab, Tb, ipt, cls, Ee, rows_pad, tid, nt, 0);  // T1 = A·X_l
+            load_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee, rows_pad,
+                      tid, nt);                           // T1 from forward
             __syncthreads();
             gemm_wgrad(Tb, Db, dWb + ((size_t)l * K + 1) * F * F, rows_pad,
                        tid);
@@ -368,6 +374,7 @@ std::vector<torch::Tensor> cheb_fwd_hip(
     TORCH_CHECK(K <= 2, "fused ChebConv kernel supports K<=2");
     const int rows_pad = round16(Ee);
     auto acts = torch::empty({B, L + 1, Ee, F}, x.options());
+    auto t1s = torch::empty({B, L, Ee, F}, x.options());
     auto lam = torch::empty({B, Ee}, x.options());
     size_t lds = sizeof(float) *
         (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F + F);
@@ -381,14 +388,15 @@ std::vector<torch::Tensor> cheb_fwd_hip(
                        x.data_ptr<float>(), W.data_ptr<float>(),
                        bias.data_ptr<float>(), ext_indptr.data_ptr<int>(),
                        ext_base.data_ptr<long>(), ext_cols.data_ptr<int>(),
-                       acts.data_ptr<float>(), lam.data_ptr<float>(),
+                       acts.data_ptr<float>(), t1s.data_ptr<float>(),
+                       lam.data_ptr<float>(),
                        B, Ee, L, K, rows_pad, (int)max_nnz, stage_csr);
-    return {lam, acts};
+    return {lam, acts, t1s};
 }
 
 std::vector<torch::Tensor> cheb_bwd_hip(
-    torch::Tensor dlam, torch::Tensor acts, torch::Tensor W,
-    torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor dlam, torch::Tensor acts, torch::Tensor t1s,
+    torch::Tensor W, torch::Tensor ext_indptr, torch::Tensor ext_base,
     torch::Tensor ext_cols, long max_nnz) {
     const int B = dlam.size(0), Ee = dlam.size(1);
     const int L = W.size(0), K = W.size(1);
@@ -405,6 +413,7 @@ std::vector<torch::Tensor> cheb_bwd_hip(
     hipLaunchKernelGGL(cheb_bwd_kernel, dim3(B), dim3(512), lds,
                        stream.stream(),
                        dlam.data_ptr<float>(), acts.data_ptr<float>(),
+                       t1s.data_ptr<float>(),
                        W.data_ptr<float>(), ext_indptr.data_ptr<int>(),
                        ext_base.data_ptr<long>(), ext_cols.data_ptr<int>(),
                        dW.data_ptr<float>(), db.data_ptr<float>(),
