@@ -94,6 +94,52 @@ DEV_INLINE void gemm_wgrad(const float* __restrict__ src,
     }
 }
 
+// ---- fused layer tile: Xb <- act(X·W0 + T·W1 + b) -----------------------
+// Safe in place: every tile's A-operands read only the tile's own rows, so
+// writing the activated output back into those rows races nothing.
+DEV_INLINE void gemm_layer_fused(float* __restrict__ Xb,
+                                 const float* __restrict__ Tb,
+                                 const float* __restrict__ Wl,
+                                 const float* __restrict__ bl,
+                                 int K, bool last, int rows_pad, int tid) {
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int nw = blockDim.x >> 6;
+    const int mtiles = rows_pad / 16;
+    const int r_in = lane & 15;
+    const int k_in = lane >> 4;
+    for (int t = wid; t < mtiles * 2; t += nw) {
+        const int mt = t >> 1;
+        const int c0 = (t & 1) * 16;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kk = 0; kk < F / 4; ++kk) {
+            const int k = kk * 4 + k_in;
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                Xb[(mt * 16 + r_in) * STRIDE + k], Wl[k * F + c0 + r_in],
+                acc, 0, 0, 0);
+        }
+        if (K > 1) {
+#pragma unroll
+            for (int kk = 0; kk < F / 4; ++kk) {
+                const int k = kk * 4 + k_in;
+                acc = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    Tb[(mt * 16 + r_in) * STRIDE + k],
+                    Wl[F * F + k * F + c0 + r_in], acc, 0, 0, 0);
+            }
+        }
+        const int col = c0 + (lane & 15);
+        const float bv = bl[col];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int row = mt * 16 + (lane >> 4) * 4 + r;
+            float y = acc[r] + bv;
+            y = last ? (y > 0.f ? y : 0.f) : (y > 0.f ? y : 0.2f * y);
+            Xb[row * STRIDE + col] = y;
+        }
+    }
+}
+
 // ---- activation tile <-> global copies (coalesced, float4 on global) -----
 // LDS rows have stride 33 (misaligned for vector LDS ops); global rows are
 // dense F=32 floats, 16B-aligned.  Task split (row, 4-col group) puts
@@ -175,8 +221,7 @@ __global__ void cheb_fwd_kernel(
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* Xb = reinterpret_cast<float*>(smem_raw);
     float* Tb = Xb + (size_t)rows_pad * STRIDE;
-    float* Yb = Tb + (size_t)rows_pad * STRIDE;
-    float* Wl = Yb + (size_t)rows_pad * STRIDE;   // K*32*32
+    float* Wl = Tb + (size_t)rows_pad * STRIDE;   // K*32*32
     float* bl = Wl + (size_t)K * F * F;           // 32
 
     const int b = blockIdx.x;
@@ -215,32 +260,15 @@ __global__ void cheb_fwd_kernel(
         for (int i = tid; i < K * F * F; i += nt)
             Wl[i] = W[((size_t)l * K) * F * F + i];
         for (int i = tid; i < F; i += nt) bl[i] = bias[l * F + i];
-        for (int i = tid; i < rows_pad * STRIDE; i += nt) Yb[i] = 0.f;
-        __syncthreads();
-
-        gemm_acc(Xb, Yb, Wl, false, rows_pad, tid);           // Y += X·W0
         __syncthreads();
         if (K > 1) {
             spmv(Xb, Tb, ipt, cls, Ee, rows_pad, tid, nt, 0); // T1 = A·X
             __syncthreads();
-            gemm_acc(Tb, Yb, Wl + F * F, false, rows_pad, tid);
-            // save T1 for the backward weight-gradient pass (skips the
-            // SpMV recompute there)
+            // save T1 for the backward weight-gradient pass
             store_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee, tid, nt);
-            __syncthreads();
         }
-        // epilogue: bias + activation → Xb (next layer input)
-        const bool last = (l == L - 1);
-        for (int r = tid; r < rows_pad; r += nt) {
-            float* y = Yb + r * STRIDE;
-            float* xo = Xb + r * STRIDE;
-            for (int c = 0; c < F; ++c) {
-                float v = y[c] + bl[c];
-                v = last ? (v > 0.f ? v : 0.f)
-                         : (v > 0.f ? v : 0.2f * v);
-                xo[c] = v;
-            }
-        }
+        // in-register tile product + activation, written back into Xb
+        gemm_layer_fused(Xb, Tb, Wl, bl, K, l == L - 1, rows_pad, tid);
         __syncthreads();
         store_acts(Xb, actsb + (size_t)(l + 1) * Ee * F, Ee, tid, nt);
         __syncthreads();
@@ -377,7 +405,7 @@ std::vector<torch::Tensor> cheb_fwd_hip(
     auto t1s = torch::empty({B, L, Ee, F}, x.options());
     auto lam = torch::empty({B, Ee}, x.options());
     size_t lds = sizeof(float) *
-        (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F + F);
+        (2 * (size_t)rows_pad * STRIDE + (size_t)K * F * F + F);
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for fused ChebConv");
     const size_t csr_bytes = sizeof(int) * ((size_t)Ee + 1 + max_nnz);
     const int stage_csr = (lds + csr_bytes <= 160 * 1024) ? 1 : 0;
