@@ -140,6 +140,43 @@ DEV_INLINE void gemm_layer_fused(float* __restrict__ Xb,
     }
 }
 
+// ---- fused backward tile: Tb <- D·W1ᵀ, Ab <- D·W0ᵀ (registers) ----------
+DEV_INLINE void gemm_dx_fused(const float* __restrict__ Db,
+                              float* __restrict__ Ab,
+                              float* __restrict__ Tb,
+                              const float* __restrict__ Wl, int K,
+                              int rows_pad, int tid) {
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int nw = blockDim.x >> 6;
+    const int mtiles = rows_pad / 16;
+    const int r_in = lane & 15;
+    const int k_in = lane >> 4;
+    for (int t = wid; t < mtiles * 2; t += nw) {
+        const int mt = t >> 1;
+        const int c0 = (t & 1) * 16;
+        f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+        f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kk = 0; kk < F / 4; ++kk) {
+            const int k = kk * 4 + k_in;
+            const float d = Db[(mt * 16 + r_in) * STRIDE + k];
+            acc0 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                d, Wl[(c0 + r_in) * F + k], acc0, 0, 0, 0);
+            if (K > 1)
+                acc1 = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                    d, Wl[F * F + (c0 + r_in) * F + k], acc1, 0, 0, 0);
+        }
+        const int col = c0 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const int row = mt * 16 + (lane >> 4) * 4 + r;
+            Ab[row * STRIDE + col] = acc0[r];
+            if (K > 1) Tb[row * STRIDE + col] = acc1[r];
+        }
+    }
+}
+
 // ---- activation tile <-> global copies (coalesced, float4 on global) -----
 // LDS rows have stride 33 (misaligned for vector LDS ops); global rows are
 // dense F=32 floats, 16B-aligned.  Task split (row, 4-col group) puts
@@ -365,16 +402,8 @@ __global__ void cheb_bwd_kernel(
         }
         if (l == 0) break;                       // features are leaves
         __syncthreads();
-        if (K > 1) {
-            // U = Db·W1ᵀ  (into Tb), then dX = Db·W0ᵀ (into Ab) + A·U
-            for (int i = tid; i < rows_pad * STRIDE; i += nt) Tb[i] = 0.f;
-            __syncthreads();
-            gemm_acc(Db, Tb, Wl + F * F, true, rows_pad, tid);
-            __syncthreads();
-        }
-        for (int i = tid; i < rows_pad * STRIDE; i += nt) Ab[i] = 0.f;
-        __syncthreads();
-        gemm_acc(Db, Ab, Wl, true, rows_pad, tid);            // dX = Db·W0ᵀ
+        // U = Db·W1ᵀ (into Tb) and dX = Db·W0ᵀ (into Ab), fused per tile
+        gemm_dx_fused(Db, Ab, Tb, Wl, K, rows_pad, tid);
         __syncthreads();
         if (K > 1) {
             spmv(Tb, Ab, ipt, cls, Ee, rows_pad, tid, nt, 1); // dX += A·U
