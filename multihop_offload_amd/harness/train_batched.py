@@ -91,6 +91,9 @@ def main(argv=None):
     ap.add_argument("--distinct", type=int, default=32)
     ap.add_argument("--T", type=int, default=1000)
     ap.add_argument("--arrival_scale", type=float, default=0.15)
+    ap.add_argument("--arrival_scales", type=str, default=None,
+                    help="comma-separated loads cycled per step (e.g. "
+                         "0.15,0.15,0.15,0.20) — overrides --arrival_scale")
     ap.add_argument("--learning_rate", type=float, default=1e-4)
     ap.add_argument("--K", type=int, default=2)
     ap.add_argument("--explore", type=float, default=0.1)
@@ -140,9 +143,11 @@ def main(argv=None):
     explore = args.explore
     t0 = time.time()
     history = []
+    loads = ([float(x) for x in args.arrival_scales.split(",")]
+             if args.arrival_scales else [args.arrival_scale])
     for step in range(1, args.steps + 1):
         engine = engines[step % len(engines)]
-        jobs = engine.sample_jobs(args.arrival_scale, gen)
+        jobs = engine.sample_jobs(loads[step % len(loads)], gen)
         for p in engine.model.parameters():
             p.grad = None
         res = engine.gnn_episode(jobs, explore=explore, gen=gen, train=True)
