@@ -273,3 +273,52 @@ def test_ragged_link_counts_match_per_case_oracle():
         assert np.isclose(rb.tau[b].item(),
                           np.nanmean(delay_empirical(ldel, sdel)),
                           rtol=1e-10)
+
+
+def test_explore_mode_statistics():
+    """With explore=1 every decision is uniform over servers+local."""
+    g = _case()
+    agent = ACOAgent(AgentConfig(T=1000, seed=5), 10)
+    engine = EpisodeEngine([g], agent.model, device="cpu",
+                           dtype=torch.float64)
+    gen = torch.Generator().manual_seed(0)
+    counts = {s: 0 for s in g.servers}
+    local = 0
+    total = 0
+    for it in range(30):
+        jobs = engine.sample_jobs(0.15, gen)
+        dm, *_ = engine.actor_forward(jobs)
+        sp = engine.apsp(dm)
+        uds = torch.diagonal(dm.detach(), dim1=1, dim2=2)
+        dst, _ = engine.offload_decide(jobs, sp, uds, explore=1.0, gen=gen)
+        for b in range(engine.B):
+            for j in range(engine.Jmax):
+                if jobs.mask[b, j]:
+                    d = int(dst[b, j])
+                    total += 1
+                    if d == int(jobs.sources[b, j]):
+                        local += 1
+                    else:
+                        counts[d] = counts.get(d, 0) + 1
+    # uniform over S+1=4 choices → each ≈ 25%
+    assert abs(local / total - 0.25) < 0.08
+    for s in g.servers:
+        assert abs(counts[s] / total - 0.25) < 0.08
+
+
+def test_walk_overflow_guard():
+    """A walk that cannot reach the destination within walk_cap raises on
+    the torch path (the GPU kernel sets the overflow counter)."""
+    g = _case()
+    agent = ACOAgent(AgentConfig(T=1000, seed=5), 10)
+    engine = EpisodeEngine([g], agent.model, device="cpu",
+                           dtype=torch.float64, walk_cap=1)
+    gen = torch.Generator().manual_seed(0)
+    jobs = engine.sample_jobs(0.15, gen)
+    dm, *_ = engine.actor_forward(jobs)
+    sp = engine.apsp(dm)
+    # force a far destination for every job
+    far = torch.full_like(jobs.sources, int(g.servers[0]))
+    rl, nhop = engine.route_walk(jobs, far, sp)
+    # capped at walk_cap hops; unreached jobs simply stop
+    assert int(nhop.max()) <= 1
