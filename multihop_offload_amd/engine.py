@@ -731,10 +731,15 @@ class EpisodeEngine:
     # ------------------------------------------------------------ episodes
     def gnn_episode(self, jobs: JobBatch, explore: float = 0.0,
                     gen: Optional[torch.Generator] = None,
-                    train: bool = True, prob: bool = False) -> EpisodeResult:
+                    train: bool = True, prob: bool = False,
+                    per_sample: bool = False) -> EpisodeResult:
         """One full GNN episode over the batch.  With ``train=True`` the
         summed per-instance actor gradients are left in ``model.param.grad``
-        (caller applies the optimizer / DP all-reduce)."""
+        (caller applies the optimizer / DP all-reduce).  With
+        ``per_sample=True`` the result additionally carries
+        ``per_sample_grads``: one gradient set per instance (the reference's
+        replay-memory unit, gnn_offloading_agent.py:141-169), extracted from
+        the per-graph dW/db partials of the fused ChebConv backward."""
         ctx = torch.enable_grad() if train else torch.no_grad()
         with ctx:
             dm, link_delay, node_delay = self.actor_forward(jobs)
@@ -745,12 +750,17 @@ class EpisodeEngine:
             self._episode_eval(jobs, dst, sp)
 
         loss_fn = loss_mse = None
+        self.last_per_sample_grads = None
         if train:
             grad_edge, loss_fn = self.critic_backward(jobs, dst, route_links,
                                                       nhop)
             grad_dist, loss_mse = self.grad_dist_matrix(
                 grad_edge, dm, unit_mtx, written)
+            self.per_sample_request = per_sample
             dm.backward(grad_dist)
+            if per_sample:
+                self.last_per_sample_grads = self._collect_per_sample_grads()
+            self.per_sample_request = False
 
         nj = jobs.mask.sum(1)
         de = torch.where(jobs.mask, delay_emp, torch.zeros_like(delay_emp))
@@ -759,6 +769,28 @@ class EpisodeEngine:
         return EpisodeResult(tau=tau, congest=congest, num_jobs=nj,
                              delay_emp=delay_emp, loss_fn=loss_fn,
                              loss_mse=loss_mse)
+
+    def _collect_per_sample_grads(self):
+        """Slice the per-graph dW/db partials stashed by ChebStackFn's
+        backward into one gradient set per instance (the reference's
+        replay-memory unit)."""
+        raw = getattr(self, "_per_sample_raw", None)
+        if raw is None:
+            raise RuntimeError(
+                "per-sample gradients need the fused ChebConv GPU path")
+        dW, db = raw
+        self._per_sample_raw = None
+        params = list(self.model.parameters())
+        L = len(params) // 2
+        out = []
+        for b in range(self.B):
+            gset = []
+            for l in range(L):
+                kw, fi, fo = params[2 * l].shape
+                gset.append(dW[b, l, :kw, :fi, :fo])
+                gset.append(db[b, l, :params[2 * l + 1].shape[0]])
+            out.append(gset)
+        return out
 
     def baseline_episode(self, jobs: JobBatch) -> EpisodeResult:
         """Greedy baseline (AdHoc_train.py:126-142), batched."""

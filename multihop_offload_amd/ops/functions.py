@@ -68,6 +68,8 @@ class ChebStackFn(torch.autograd.Function):
             dW, db = ext.cheb_bwd(dlam.contiguous(), acts, ctx.t1s, Wp,
                                   eng.k_ext_indptr, eng.k_ext_base,
                                   eng.k_ext_cols, eng.k_ext_max_nnz)
+        if getattr(eng, "per_sample_request", False):
+            eng._per_sample_raw = (dW, db)   # per-graph partials, pre-sum
         dW = dW.sum(dim=0)            # (L,K,32,32) summed over graphs
         db = db.sum(dim=0)            # (L,32)
         grads = []

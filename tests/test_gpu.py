@@ -319,3 +319,40 @@ def test_reference_checkpoint_in_engine_gpu():
     rl = engine.local_episode(jobs)
     # their trained policy collapses to local computing on these loads
     assert torch.allclose(res.tau, rl.tau, rtol=1e-3)
+
+
+@needs_gpu
+def test_per_sample_gradients_match_oracle():
+    """engine.gnn_episode(per_sample=True) must reproduce the reference's
+    per-instance gradient sets (oracle agent, fp32)."""
+    from multihop_offload_amd import ACOAgent, AdhocCloudEnv
+    from multihop_offload_amd.agent import AgentConfig
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.graphs import JobInstance
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from tests.test_engine import _case, _jobbatch_from, _wake
+
+    g1, g2 = _case(seed=31), _case(seed=33)
+    agent = ACOAgent(AgentConfig(T=1000, seed=5, dtype="float32"), 10)
+    _wake(agent.model)
+    model_g = ChebConvStack(K=2, dtype=torch.float32, seed=5)
+    with torch.no_grad():
+        for pc, pg in zip(agent.model.parameters(), model_g.parameters()):
+            pg.copy_(pc)
+    engine = EpisodeEngine([g1, g2], model_g, device="cuda",
+                           dtype=torch.float32)
+    j1 = JobInstance.sample(g1.mobile_nodes, 0.15, np.random.RandomState(1))
+    j2 = JobInstance.sample(g2.mobile_nodes, 0.15, np.random.RandomState(2))
+    res = engine.gnn_episode(_jobbatch_from(engine, [j1, j2]), train=True,
+                             per_sample=True)
+    psg = engine.last_per_sample_grads
+    assert len(psg) == 2
+    for b, (g, j) in enumerate(((g1, j1), (g2, j2))):
+        env = AdhocCloudEnv(g)
+        env.set_jobs(j)
+        agent.forward_backward(env, 0.0, np.random.RandomState(0))
+        oracle = agent.memory[-1][0]
+        for got, want in zip(psg[b], oracle):
+            a = got.cpu().numpy()
+            w = want.cpu().numpy()
+            assert np.abs(a - w).max() / max(np.abs(w).max(), 1e-6) < 1e-2
