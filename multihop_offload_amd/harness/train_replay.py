@@ -46,7 +46,8 @@ def main(argv=None):
     ap.add_argument("--distinct", type=int, default=64)
     ap.add_argument("--T", type=int, default=1000)
     ap.add_argument("--arrival_scale", type=float, default=0.15)
-    ap.add_argument("--learning_rate", type=float, default=1e-4)
+    ap.add_argument("--learning_rate", type=float, default=1e-6,
+                    help="the reference train.sh value; each replay applies\n                    ~replay_batch sequential Adam steps, so keep it small")
     ap.add_argument("--K", type=int, default=2)
     ap.add_argument("--replay_batch", type=int, default=100)
     ap.add_argument("--replay_every", type=int, default=1)
