@@ -36,7 +36,7 @@ from typing import List, Optional
 import numpy as np
 import torch
 
-from .env import AdhocCloudEnv, apsp, delay_empirical
+from .env import AdhocCloudEnv, apsp
 from .graphs import CaseGraph, JobInstance
 from .models.chebconv import ChebConvStack
 from .queueing import ConflictCSR, actor_delays, delay_matrix, delay_with_fallback, fixed_point_mu

@@ -24,7 +24,7 @@ import numpy as np
 import scipy.io as sio
 import scipy.sparse as sp
 
-from .graphs import CaseGraph, build_connectivity
+from .graphs import build_connectivity
 
 GRAPH_SIZES = [20, 30, 40, 50, 60, 70, 80, 90, 100, 110]
 

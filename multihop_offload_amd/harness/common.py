@@ -8,14 +8,13 @@ from __future__ import annotations
 
 import argparse
 import os
-import time
 from typing import Optional
 
 import numpy as np
 
 from ..agent import ACOAgent, AgentConfig
 from ..env import AdhocCloudEnv, apsp, delay_empirical
-from ..graphs import CaseGraph, JobInstance
+from ..graphs import CaseGraph
 
 
 def build_parser(default_datapath="../data_100") -> argparse.ArgumentParser:

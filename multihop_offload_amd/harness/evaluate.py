@@ -15,7 +15,6 @@ import argparse
 import json
 import os
 
-import numpy as np
 import torch
 
 from ..engine import EpisodeEngine

@@ -11,7 +11,7 @@ the right shape for xGMI.
 from __future__ import annotations
 
 import os
-from typing import Iterable, Optional
+from typing import Iterable
 
 import torch
 import torch.distributed as dist
