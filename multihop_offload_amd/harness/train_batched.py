@@ -107,6 +107,10 @@ def main(argv=None):
     ap.add_argument("--workers", type=int, default=8)
     ap.add_argument("--lr_decay_at", type=int, default=0,
                     help="step after which lr is multiplied by 0.1")
+    ap.add_argument("--guard_every", type=int, default=1000,
+                    help="divergence guard period (0 disables): if the "
+                         "running tau explodes past 5x the best seen, roll "
+                         "back to the best parameters and cut lr 3x")
     ap.add_argument("--init_scale", type=float, default=0.01,
                     help="shrink initial weights (wakes the output ReLU)")
     args = ap.parse_args(argv)
@@ -143,6 +147,13 @@ def main(argv=None):
     explore = args.explore
     t0 = time.time()
     history = []
+    # divergence guard state (the semi-analytic gradient is unstable near
+    # the 1/(mu-lambda) poles; see ROUND2.md — rollback + lr cut recovers)
+    import copy as _copy
+    best_tau = float("inf")
+    best_params = None
+    best_opt = None
+    rollbacks = 0
     loads = ([float(x) for x in args.arrival_scales.split(",")]
              if args.arrival_scales else [args.arrival_scale])
     for step in range(1, args.steps + 1):
@@ -167,6 +178,28 @@ def main(argv=None):
         if args.lr_decay_at and step == args.lr_decay_at:
             for group in opt.param_groups:
                 group["lr"] *= 0.1
+        if args.guard_every and step % args.guard_every == 0:
+            tau_now = float(torch.nanmean(res.tau))
+            if tau_now < best_tau:
+                best_tau = tau_now
+                best_params = [p.detach().clone()
+                               for p in engine.model.parameters()]
+                best_opt = _copy.deepcopy(opt.state_dict())
+            elif best_params is not None and \
+                    tau_now > 5.0 * max(best_tau, 20.0):
+                with torch.no_grad():
+                    for p, bp in zip(engine.model.parameters(), best_params):
+                        p.copy_(bp)
+                opt.load_state_dict(best_opt)
+                for group in opt.param_groups:
+                    group["lr"] /= 3.0
+                rollbacks += 1
+                if rank == 0:
+                    print(json.dumps({"step": step, "rollback": rollbacks,
+                                      "tau_now": tau_now,
+                                      "best_tau": best_tau,
+                                      "lr": opt.param_groups[0]["lr"]}),
+                          flush=True)
 
         if step % args.log_every == 0 and rank == 0:
             tau = float(torch.nanmean(res.tau))
