@@ -218,6 +218,12 @@ def main(argv=None):
                          os.path.join(actor_dir,
                                       f"cp-{step // args.save_every:04d}.ckpt"))
     if rank == 0:
+        # ship the best-seen parameters (the guard may have kept training
+        # through late instability)
+        if best_params is not None:
+            with torch.no_grad():
+                for p, bp in zip(engine.model.parameters(), best_params):
+                    p.copy_(bp)
         ckpt_io.save(engine.model, os.path.join(actor_dir, "cp-9999.ckpt"))
         with open(os.path.join(args.model_root,
                                f"train_history_{args.training_set}.json"),
