@@ -12,8 +12,10 @@ on CPU (fp64) where it is tested for exact agreement with the oracle
 
 Batching layout:
   * per-graph dense tensors are stacked: (B, N, N) / (B, E) / (B, Ē);
-    all graphs in a batch share N (and hence E for BA(m)); ragged
-    conflict structure is flat block-diagonal CSR over B·E (resp. B·Ē);
+    all graphs in a batch share N; link counts MAY differ (ragged E — E is
+    the batch max, per-graph counts in E_arr, virtual extended edges
+    renumbered to start at E); conflict structure is flat block-diagonal
+    CSR over B·E (resp. B·Ē); horizons T may differ per graph;
   * job instances are padded to Jmax with a validity mask; padded jobs have
     zero rate and are excluded from routes, losses and metrics;
   * the per-instance actor gradient sets of the reference's replay memory

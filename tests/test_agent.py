@@ -220,3 +220,19 @@ def test_param_count_matches_reference_at_k1():
     agent = _agent(K=1)
     n = sum(p.numel() for p in agent.model.parameters())
     assert n == 3361
+
+
+def test_latest_checkpoint_manifest_priority(tmp_path):
+    """The manifest entry wins over lexicographic file order (reference
+    tf.train.latest_checkpoint protocol)."""
+    from multihop_offload_amd.utils.checkpoint import latest_checkpoint
+    import numpy as np
+    d = tmp_path / "model_ChebConv_X_a5_c5_ACO_agent"
+    d.mkdir()
+    for name in ("cp-0001.ckpt.npz", "cp-0005.ckpt.npz"):
+        np.savez(str(d / name)[:-4] + ".npz")
+    (d / "checkpoint").write_text(
+        'model_checkpoint_path: "cp-0001.ckpt"\n')
+    assert latest_checkpoint(str(d)).endswith("cp-0001.ckpt")
+    (d / "checkpoint").unlink()
+    assert latest_checkpoint(str(d)).endswith("cp-0005.ckpt")

@@ -97,3 +97,43 @@ def test_actor_delays_pipeline_gradflow(small_case):
     (ld.sum() + nd.sum()).backward()
     assert torch.isfinite(lam_l.grad).all()
     assert torch.isfinite(lam_n.grad).all()
+
+
+def test_fixed_point_batched_equals_per_graph(small_case):
+    """Block-diagonal flat batch == per-graph runs."""
+    import numpy as np
+    from multihop_offload_amd.graphs import CaseGraph
+    g1 = small_case
+    rng = np.random.RandomState(9)
+    g2 = CaseGraph(20, seed=9, gtype="ba")
+    g2.links_init(40.0, rng=rng)
+    E = g1.num_links
+    assert g2.num_links == E
+    lam = torch.tensor(rng.uniform(0, 40, 2 * E))
+    # flat batch CSR
+    rows = np.concatenate([
+        np.repeat(np.arange(E), np.diff(g1.conf_indptr)),
+        np.repeat(np.arange(E), np.diff(g2.conf_indptr)) + E])
+    cols = np.concatenate([g1.conf_indices, g2.conf_indices + E])
+    conf = ConflictCSR.__new__(ConflictCSR)
+    conf.row = torch.tensor(rows)
+    conf.col = torch.tensor(cols)
+    conf.n = 2 * E
+    rates = torch.tensor(np.concatenate([g1.link_rates, g2.link_rates]))
+    cf = torch.tensor(np.concatenate([g1.cf_degs, g2.cf_degs]))
+    mu_flat = fixed_point_mu(lam, rates, cf, conf)
+    for i, g in enumerate((g1, g2)):
+        c = ConflictCSR(g.conf_indptr, g.conf_indices)
+        mu = fixed_point_mu(lam[i * E:(i + 1) * E],
+                            torch.tensor(g.link_rates),
+                            torch.tensor(g.cf_degs), c)
+        assert torch.allclose(mu_flat[i * E:(i + 1) * E], mu)
+
+
+def test_delay_fallback_tensor_T():
+    lam = torch.tensor([10.0, 10.0], dtype=torch.float64)
+    mu = torch.tensor([5.0, 5.0], dtype=torch.float64)
+    T = torch.tensor([700.0, 1000.0], dtype=torch.float64)
+    d = delay_with_fallback(lam, mu, T, 101.0)
+    assert np.isclose(d[0].item(), 700.0 * 10 / (101 * 5))
+    assert np.isclose(d[1].item(), 1000.0 * 10 / (101 * 5))

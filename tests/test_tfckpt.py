@@ -46,3 +46,11 @@ def test_load_into_model_and_forward(small_case, jobs_for):
     m2 = ChebConvStack(K=2, dtype=torch.float64)
     with pytest.raises(ValueError):
         load_reference_weights(m2, PREFIX)
+
+
+def test_bad_magic_rejected(tmp_path):
+    from multihop_offload_amd.utils.tfckpt import read_bundle
+    p = tmp_path / "fake.ckpt"
+    (tmp_path / "fake.ckpt.index").write_bytes(b"\x00" * 64)
+    with pytest.raises(ValueError):
+        read_bundle(str(p))
