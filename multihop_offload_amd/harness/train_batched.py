@@ -205,9 +205,13 @@ def main(argv=None):
             tau = float(torch.nanmean(res.tau))
             congest = int(res.congest.sum())
             njobs = int(res.num_jobs.sum())
+            gnorm = float(sum((p.grad ** 2).sum()
+                              for p in engine.model.parameters()
+                              if p.grad is not None) ** 0.5)
             rec = {"step": step, "tau": tau,
                    "congest_ratio": congest / max(njobs, 1),
                    "loss_fn": float(res.loss_fn), "loss_mse": float(res.loss_mse),
+                   "grad_norm": gnorm,
                    "explore": explore,
                    "eps_per_sec": engine.B * world * step
                    / (time.time() - t0)}
