@@ -87,3 +87,27 @@ def test_forward_gcn_variant(small_case, jobs_for):
     agent2 = ACOAgent(AgentConfig(seed=0), 10)
     with pytest.raises(ValueError):
         agent2.forward_gcn(small_case, jobs_for)
+
+
+def test_dispatch_fails_loudly_without_extension(monkeypatch):
+    """GPU execution must never silently fall back to eager torch when the
+    native extension is missing."""
+    from multihop_offload_amd.ops import dispatch
+    monkeypatch.setattr(dispatch, "_HIP_EXT", None)
+    monkeypatch.setattr(dispatch, "_HIP_TRIED", True)
+    with pytest.raises(RuntimeError, match="HIP extension not built"):
+        dispatch.require_hip()
+
+
+def test_force_torch_escape_hatch(monkeypatch):
+    """MHO_FORCE_TORCH=1 routes floyd_warshall to the torch reference even
+    for CUDA tensors (debugging aid) — CPU tensors always use torch."""
+    import torch as _t
+    from multihop_offload_amd.ops import dispatch
+    monkeypatch.setenv("MHO_FORCE_TORCH", "1")
+    w = _t.full((1, 4, 4), float("inf"), dtype=_t.float64)
+    for i in range(4):
+        w[0, i, i] = 0
+        w[0, i, (i + 1) % 4] = w[0, (i + 1) % 4, i] = 1.0
+    d = dispatch.floyd_warshall(w)
+    assert d[0, 0, 2] == 2.0
