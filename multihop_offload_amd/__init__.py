@@ -17,5 +17,5 @@ designed MI355X-first:
 __version__ = "0.1.0"
 
 from .graphs import CaseGraph, JobInstance  # noqa: F401
-from .env import AdhocCloudEnv              # noqa: F401
+from .env import AdhocCloudEnv, AdhocCloud  # noqa: F401
 from .agent import ACOAgent                 # noqa: F401

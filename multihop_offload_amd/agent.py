@@ -149,6 +149,24 @@ class ACOAgent:
                           g.num_nodes)
         return dm
 
+    # -- reference state/predict API (gnn_offloading_agent.py:134-154) ----------
+    def makestate(self, g: CaseGraph, jobs: JobInstance):
+        """Bundle the GNN inputs like the reference's ``makestate``; the
+        support entry is the case itself (its CSR operator pair is cached
+        per case by ``_support``)."""
+        return {"node_features": g.ext.features(jobs), "support": g}
+
+    def predict(self, state):
+        """Run the actor GNN on a ``makestate`` bundle (reference
+        ``predict``/``act``): returns the raw (Ē, 1) head output."""
+        g = state["support"]
+        support, _ = self._support(g)
+        x = self._t(np.asarray(state["node_features"]))
+        return self.model(x, support)
+
+    def act(self, state):
+        return self.predict(state)
+
     # -- forward + environment (reference :278-291) -----------------------------
     def _env_step(self, env: AdhocCloudEnv, explore: float,
                   rng: Optional[np.random.RandomState] = None):

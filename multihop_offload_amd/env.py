@@ -234,3 +234,72 @@ def delay_empirical(link_delay: np.ndarray, server_delay: np.ndarray) -> np.ndar
     """Per-job total delay: nansum over links + nansum over servers
     (``AdHoc_train.py:140,153``)."""
     return np.nansum(link_delay, axis=0) + np.nansum(server_delay, axis=0)
+
+
+class AdhocCloud(CaseGraph):
+    """Drop-in facade with the reference's ``AdhocCloud`` API
+    (``offloading_v3.py:29-550``): constructor accepting a graph family or
+    a ``.mat`` path as ``gtype``, imperative job management, and the
+    decision/evaluation methods delegating to :class:`AdhocCloudEnv`.
+
+    New code should use :class:`~multihop_offload_amd.graphs.CaseGraph` +
+    :class:`AdhocCloudEnv` (or the batched engine) directly; this class
+    exists so reference-shaped scripts port without edits.
+    """
+
+    def __init__(self, num_nodes, t_max=1000, seed=3, m=2, pos=None,
+                 cf_radius=0.0, gtype="ba", trace=False):
+        if ".mat" in str(gtype):
+            base = CaseGraph.from_mat(gtype, t_max=t_max, cf_radius=cf_radius)
+            self.__dict__.update(base.__dict__)   # adopt graph + roles/bws
+        else:
+            super().__init__(num_nodes, t_max=t_max, seed=seed, m=m,
+                             gtype=gtype, pos=pos, cf_radius=cf_radius)
+        self.trace = trace
+        self._env = AdhocCloudEnv(self)
+        self._job_list = []
+
+    # -- job management (offloading_v3.py:243-250) ----------------------------
+    def add_job(self, src, rate=0.1, ul=100, dl=1):
+        self._job_list.append((int(src), float(rate), float(ul), float(dl)))
+        self._sync_jobs()
+
+    def clear_all_jobs(self):
+        self._job_list = []
+        self._env.jobs = None
+        self._env.flows = []
+
+    @property
+    def num_jobs(self):
+        return len(self._job_list)
+
+    def _sync_jobs(self):
+        a = np.asarray(self._job_list, dtype=np.float64).reshape(-1, 4)
+        self._env.set_jobs(JobInstance(
+            sources=a[:, 0].astype(np.int64), rates=a[:, 1],
+            ul=a[:, 2], dl=a[:, 3]))
+
+    @property
+    def flows(self):
+        return self._env.flows
+
+    # -- decision / evaluation delegates --------------------------------------
+    def graph_expand(self):
+        return self.ext
+
+    def dmtx_baseline(self):
+        return self._env.dmtx_baseline()
+
+    def local_compute(self, unit_delay_servers):
+        return self._env.local_compute(unit_delay_servers)
+
+    def offloading(self, spmtx_in, hpmtx=None, explore=0.0, prob=False):
+        if hpmtx is None:
+            hpmtx = self.sp_hop
+        return self._env.offloading(spmtx_in, hpmtx, explore, prob)
+
+    def routing(self, src, dst, spmtx):
+        return self._env.routing(int(src), int(dst), spmtx)
+
+    def run(self):
+        return self._env.run()

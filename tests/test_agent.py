@@ -236,3 +236,21 @@ def test_latest_checkpoint_manifest_priority(tmp_path):
     assert latest_checkpoint(str(d)).endswith("cp-0001.ckpt")
     (d / "checkpoint").unlink()
     assert latest_checkpoint(str(d)).endswith("cp-0005.ckpt")
+
+
+def test_makestate_predict_act(small_case, jobs_for):
+    """Reference state API (gnn_offloading_agent.py:134-154): predict(act)
+    on a makestate bundle equals the model applied to the extended
+    features."""
+    from multihop_offload_amd.agent import AgentConfig
+    agent = ACOAgent(AgentConfig(seed=0), 10)
+    state = agent.makestate(small_case, jobs_for)
+    out = agent.predict(state)
+    Ee = small_case.ext.num_edges_ext
+    assert out.shape == (Ee, 1)
+    assert torch.equal(agent.act(state), out)
+    # same lambda the full forward pass consumes
+    dm, link_delay, node_delay = agent.forward(small_case, jobs_for)
+    E = small_case.num_links
+    assert torch.isfinite(link_delay).all()
+    assert out[:E, 0].shape == link_delay.shape

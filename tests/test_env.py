@@ -206,3 +206,66 @@ def test_greedy_route_terminates_and_is_valid(small_case):
             assert route[0] == src and route[-1] == dst and nh == len(route) - 1
             for a, b in zip(route[:-1], route[1:]):
                 assert g.adj[a, b] == 1
+
+
+class TestAdhocCloudFacade:
+    """Drop-in ``AdhocCloud`` facade: reference-shaped API on one object."""
+
+    def _make(self):
+        from multihop_offload_amd import AdhocCloud
+        rng = np.random.RandomState(42)
+        net = AdhocCloud(20, t_max=1000, seed=7, gtype="ba")
+        net.links_init(50.0, rng=rng)
+        net.add_relay(0)
+        net.add_relay(1)
+        for s in (2, 3, 4):
+            net.add_server(s, 300.0)
+        for n in range(5, 20):
+            net.set_mobile_bw(n, 10.0)
+        return net
+
+    def test_facade_matches_env(self, small_case, jobs_for):
+        from multihop_offload_amd.env import AdhocCloudEnv, apsp
+        net = self._make()
+        for s, r, ul, dl in zip(jobs_for.sources, jobs_for.rates,
+                                jobs_for.ul, jobs_for.dl):
+            net.add_job(s, r, ul, dl)
+        assert net.num_jobs == jobs_for.num_jobs
+
+        env = AdhocCloudEnv(small_case)
+        env.set_jobs(jobs_for)
+        _, dlist, dproc = env.dmtx_baseline()
+        dmtx_f, dlist_f, dproc_f = net.dmtx_baseline()
+        np.testing.assert_allclose(dlist_f, dlist)
+        np.testing.assert_allclose(dproc_f, dproc)
+
+        sp = apsp(small_case, dlist)
+        np.fill_diagonal(sp, np.where(dproc > 0, dproc, small_case.T))
+        dec_f, del_f = net.offloading(sp)        # hpmtx defaults to sp_hop
+        dec_e, del_e = env.offloading(sp, small_case.sp_hop)
+        np.testing.assert_array_equal(dec_f, dec_e)
+        np.testing.assert_allclose(del_f, del_e)
+
+        ld_f, sd_f, um_f = net.run()
+        ld_e, sd_e, um_e = env.run()
+        np.testing.assert_allclose(ld_f, ld_e, equal_nan=True)
+        np.testing.assert_allclose(sd_f, sd_e, equal_nan=True)
+        np.testing.assert_allclose(um_f, um_e, equal_nan=True)
+
+        net.clear_all_jobs()
+        assert net.num_jobs == 0 and net.flows == []
+
+    def test_facade_from_mat(self, tmp_path):
+        import glob
+        mats = glob.glob("data_samples/**/*.mat", recursive=True)
+        if not mats:
+            pytest.skip("no committed .mat samples")
+        from multihop_offload_amd import AdhocCloud
+        net = AdhocCloud(0, gtype=mats[0])
+        assert net.num_nodes > 0 and len(net.servers) > 0
+        net.links_init(net.mat_link_rate, rng=np.random.RandomState(0))
+        net.add_job(int(net.mobile_nodes[0]), 0.1)
+        route, nhop = net.routing(int(net.mobile_nodes[0]),
+                                  int(net.servers[0]), net.sp_hop)
+        assert route[0] == net.mobile_nodes[0] and route[-1] == net.servers[0]
+        assert net.graph_expand() is net.ext
