@@ -298,8 +298,31 @@ class AdhocCloud(CaseGraph):
             hpmtx = self.sp_hop
         return self._env.offloading(spmtx_in, hpmtx, explore, prob)
 
-    def routing(self, src, dst, spmtx):
-        return self._env.routing(int(src), int(dst), spmtx)
+    def routing(self, flow, spmtx, *rest):
+        """Reference form ``routing(flow, spmtx)``; also accepts
+        ``routing(src, dst, spmtx)`` with plain node ids."""
+        if rest:
+            src, dst, spmtx = int(flow), int(spmtx), rest[0]
+        else:
+            src, dst = int(flow.src), int(flow.dst)
+        return self._env.routing(src, dst, spmtx)
 
     def run(self):
         return self._env.run()
+
+    def plot_routes(self, link_delays, node_delays, opt, with_labels=True,
+                    fig_dir="fig"):
+        from .utils.plotting import plot_routes as _pr
+        return _pr(self, self._env, link_delays, node_delays, opt,
+                   fig_dir=fig_dir, with_labels=with_labels)
+
+    def plot_metrics(self, opt, T=None, seed=0, fig_dir="fig"):
+        """Run the per-timeslot tracer over the current flows and save the
+        reference metrics figure (offloading_v3.py:588-607)."""
+        from .sim.timeslot import simulate
+        from .utils.plotting import plot_metrics as _pm
+        _, _, trace = simulate(self, self._env.jobs, self._env.flows,
+                               T=int(T or self.T), seed=seed, trace=True)
+        _pm(trace, self, opt, fig_dir=fig_dir)
+        return (trace["arrivals"], trace["pkts_in_network"],
+                trace["departures"])

@@ -41,8 +41,12 @@ class _Task:
 
 
 def simulate(g: CaseGraph, jobs: JobInstance, flows: List[Flow],
-             T: int = 2000, seed: int = 0, warmup: int = 0):
-    """Returns (mean_delay_per_job (J,), completed_counts (J,))."""
+             T: int = 2000, seed: int = 0, warmup: int = 0,
+             trace: bool = False):
+    """Returns (mean_delay_per_job (J,), completed_counts (J,)); with
+    ``trace=True`` also a dict of per-slot totals (arrivals, sink
+    departures, packets in network — the reference ``plot_metrics``
+    series, offloading_v3.py:588-607)."""
     rng = np.random.RandomState(seed)
     E, J = g.num_links, jobs.num_jobs
 
@@ -60,6 +64,10 @@ def simulate(g: CaseGraph, jobs: JobInstance, flows: List[Flow],
     link_q: List[List[_Task]] = [[] for _ in range(E)]
     node_q: List[List[_Task]] = [[] for _ in range(g.num_nodes)]
     delays = [[] for _ in range(J)]
+    tr_arr = np.zeros(T, dtype=np.int64)
+    tr_dep = np.zeros(T, dtype=np.int64)
+    tr_net = np.zeros(T, dtype=np.int64)
+    in_network = 0
 
     def enqueue(task: _Task, t: int):
         while task.stage < len(legs[task.job]):
@@ -76,7 +84,12 @@ def simulate(g: CaseGraph, jobs: JobInstance, flows: List[Flow],
         for j in range(J):
             for _ in range(rng.poisson(jobs.rates[j])):
                 task = _Task(j, t)
+                tr_arr[t] += 1
+                in_network += 1
                 enqueue(task, t)
+                if task.done_at >= 0:           # zero-length leg sequence
+                    in_network -= 1
+                    tr_dep[t] += 1
         # link service with contention sharing
         busy = np.array([1 if q else 0 for q in link_q])
         nb_busy = np.zeros(E)
@@ -115,7 +128,14 @@ def simulate(g: CaseGraph, jobs: JobInstance, flows: List[Flow],
                     break
         for task in finished:
             enqueue(task, t + 1)
+            if task.done_at >= 0:
+                in_network -= 1
+                tr_dep[min(t + 1, T - 1)] += 1
+        tr_net[t] = in_network
 
     mean_delay = np.array([np.mean(d) if d else np.nan for d in delays])
     counts = np.array([len(d) for d in delays])
+    if trace:
+        return mean_delay, counts, {"arrivals": tr_arr, "departures": tr_dep,
+                                    "pkts_in_network": tr_net}
     return mean_delay, counts

@@ -68,3 +68,29 @@ def plot_routes(case, env, link_delays, node_delays, opt, fig_dir="fig",
     plt.savefig(name, dpi=300, bbox_inches="tight")
     plt.close()
     return name
+
+
+def plot_metrics(trace, case, opt, fig_dir="fig"):
+    """Reference ``AdhocCloud.plot_metrics`` (offloading_v3.py:588-607):
+    per-timeslot exogenous arrivals, sink departures, and packets in the
+    network, from a ``sim.timeslot.simulate(..., trace=True)`` trace."""
+    import matplotlib
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    plt.plot(trace["arrivals"])
+    plt.plot(trace["departures"])
+    plt.plot(trace["pkts_in_network"])
+    plt.suptitle("Departures, Arrivals, and Current amount pkts in network")
+    plt.xlabel("T")
+    plt.ylabel("the number of packages")
+    plt.legend(["Exogenous arrivals", "Sink departures", "Pkts in network"],
+               loc="upper right")
+    os.makedirs(fig_dir, exist_ok=True)
+    name = os.path.join(
+        fig_dir, "flow_packets_arrivals_per_timeslot_seed_{}_nodes_{}_{}"
+        "_cf{:.1f}_opt_{}.png".format(case.seed, case.num_nodes, case.gtype,
+                                      case.cf_radius, opt))
+    plt.savefig(name, dpi=300)
+    plt.close()
+    return name
