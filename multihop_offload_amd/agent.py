@@ -264,6 +264,21 @@ class ACOAgent:
             float(loss_fn.detach()), loss_mse
 
     # -- replay (reference :141-169) --------------------------------------------
+    # -- scalar logging (reference log_init/log_scalar, :455-469) ---------------
+    def log_init(self, logdir="logs"):
+        import datetime
+        import os
+        stamp = datetime.datetime.now().strftime("%Y%m%d-%H%M%S")
+        base = os.path.join(logdir, "gradient_tape", stamp)
+        self.train_summary_writer = _ScalarLogger(
+            os.path.join(base, "train.jsonl"))
+        self.test_summary_writer = _ScalarLogger(
+            os.path.join(base, "test.jsonl"))
+
+    def log_scalar(self, name, variable, step, test=False):
+        w = self.test_summary_writer if test else self.train_summary_writer
+        w.scalar(name, variable, step)
+
     def memorize(self, grads, loss, reward):
         self.memory.append((grads, loss, reward))
 
@@ -291,3 +306,22 @@ class ACOAgent:
         if self.epsilon > self.cfg.epsilon_min:
             self.epsilon *= self.cfg.epsilon_decay
         return float(np.nanmean(losses))
+
+
+class _ScalarLogger:
+    """JSONL scalar logger — the MI355X-native stand-in for the reference's
+    TF summary writers (``gnn_offloading_agent.py:455-469``; commented out in
+    its harnesses).  One line per scalar: {"name", "value", "step"}."""
+
+    def __init__(self, path: str):
+        import os
+        os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
+        self._fh = open(path, "a", buffering=1)
+
+    def scalar(self, name: str, value, step: int):
+        import json
+        self._fh.write(json.dumps(
+            {"name": name, "value": float(value), "step": int(step)}) + "\n")
+
+    def close(self):
+        self._fh.close()

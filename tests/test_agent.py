@@ -254,3 +254,17 @@ def test_makestate_predict_act(small_case, jobs_for):
     E = small_case.num_links
     assert torch.isfinite(link_delay).all()
     assert out[:E, 0].shape == link_delay.shape
+
+
+def test_log_scalar_jsonl(tmp_path):
+    import json
+    from multihop_offload_amd.agent import AgentConfig
+    agent = ACOAgent(AgentConfig(seed=0), 10)
+    agent.log_init(logdir=str(tmp_path))
+    agent.log_scalar("loss", 1.5, step=3)
+    agent.log_scalar("tau", 2.0, step=4, test=True)
+    logs = sorted((tmp_path / "gradient_tape").rglob("*.jsonl"))
+    assert len(logs) == 2
+    row = json.loads(open([l for l in logs if l.name == "train.jsonl"][0]
+                          ).readline())
+    assert row == {"name": "loss", "value": 1.5, "step": 3}
