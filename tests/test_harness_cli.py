@@ -43,3 +43,18 @@ def test_adhoc_train_then_test_cli(tmp_path):
     assert len(tcsvs) == 1
     tdf = pd.read_csv(tcsvs[0])
     assert "Algo" in tdf.columns and set(tdf["Algo"]) >= {"baseline", "local"}
+
+
+def test_demo_main(tmp_path):
+    """The reference demo path (offloading_v3.main, stale/crashing as
+    shipped) runs end-to-end here, including the metrics trace figure."""
+    from multihop_offload_amd.harness import demo
+    dec, est, emp = demo.main([
+        "1", "--fig_dir", str(tmp_path / "fig"),
+        "--out", str(tmp_path / "out"), "--trace"])
+    assert len(dec) == len(est) == len(emp) == 5
+    assert (emp > 0).all()
+    figs = os.listdir(tmp_path / "fig")
+    assert any("flow_routes" in f for f in figs)
+    assert any("flow_packets" in f for f in figs)
+    assert os.listdir(tmp_path / "out")
