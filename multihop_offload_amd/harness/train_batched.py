@@ -79,6 +79,35 @@ def build_training_cases(n_nodes, batch, distinct, T, seed, gtype="ba",
     return cases
 
 
+def _dist_mean(x: float, world: int, device) -> float:
+    """Average a scalar across data-parallel ranks so control-flow
+    decisions (divergence guard, eval-based selection) are identical on
+    every rank — per-rank decisions would silently de-synchronise the
+    replicated parameters."""
+    if world > 1 and torch.distributed.is_initialized():
+        dev = device if torch.distributed.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([x], dtype=torch.float32, device=dev)
+        torch.distributed.all_reduce(t)
+        return float(t.item()) / world
+    return float(x)
+
+
+def evaluate_policy(engines, loads, seed: int, rounds: int = 4) -> float:
+    """Held-out evaluation: mean per-job tau of the greedy policy
+    (explore=0, no gradients) over fresh job draws from a FIXED seed, so
+    successive calls during training are comparable."""
+    taus = []
+    with torch.no_grad():
+        for engine in engines:
+            gen = torch.Generator(device=engine.device)
+            gen.manual_seed(seed)
+            for r in range(rounds):
+                jobs = engine.sample_jobs(loads[r % len(loads)], gen)
+                res = engine.gnn_episode(jobs, train=False)
+                taus.append(res.tau.flatten())
+    return float(torch.nanmean(torch.cat(taus)))
+
+
 def main(argv=None):
     ap = argparse.ArgumentParser()
     ap.add_argument("--steps", type=int, default=2000)
@@ -113,6 +142,13 @@ def main(argv=None):
                          "back to the best parameters and cut lr 3x")
     ap.add_argument("--init_scale", type=float, default=0.01,
                     help="shrink initial weights (wakes the output ReLU)")
+    ap.add_argument("--eval_every", type=int, default=0,
+                    help="held-out greedy evaluation period (0 disables); "
+                         "the final checkpoint is then the best-by-eval-tau "
+                         "parameters, a stronger selection signal than the "
+                         "training tau the guard tracks")
+    ap.add_argument("--eval_rounds", type=int, default=4)
+    ap.add_argument("--eval_seed", type=int, default=12345)
     args = ap.parse_args(argv)
 
     rank, world = dp.init_from_env()
@@ -154,6 +190,8 @@ def main(argv=None):
     best_params = None
     best_opt = None
     rollbacks = 0
+    best_eval_tau = float("inf")
+    best_eval_params = None
     loads = ([float(x) for x in args.arrival_scales.split(",")]
              if args.arrival_scales else [args.arrival_scale])
     for step in range(1, args.steps + 1):
@@ -178,8 +216,23 @@ def main(argv=None):
         if args.lr_decay_at and step == args.lr_decay_at:
             for group in opt.param_groups:
                 group["lr"] *= 0.1
+        if args.eval_every and step % args.eval_every == 0:
+            eval_tau = _dist_mean(
+                evaluate_policy(engines, loads, args.eval_seed,
+                                args.eval_rounds),
+                world, engine.device)
+            if eval_tau < best_eval_tau:
+                best_eval_tau = eval_tau
+                best_eval_params = [p.detach().clone()
+                                    for p in engine.model.parameters()]
+            if rank == 0:
+                rec = {"step": step, "eval_tau": eval_tau,
+                       "best_eval_tau": best_eval_tau}
+                history.append(rec)
+                print(json.dumps(rec), flush=True)
         if args.guard_every and step % args.guard_every == 0:
-            tau_now = float(torch.nanmean(res.tau))
+            tau_now = _dist_mean(float(torch.nanmean(res.tau)),
+                                 world, engine.device)
             if tau_now < best_tau:
                 best_tau = tau_now
                 best_params = [p.detach().clone()
@@ -222,11 +275,12 @@ def main(argv=None):
                          os.path.join(actor_dir,
                                       f"cp-{step // args.save_every:04d}.ckpt"))
     if rank == 0:
-        # ship the best-seen parameters (the guard may have kept training
-        # through late instability)
-        if best_params is not None:
+        # ship the best-seen parameters: held-out-eval selection when
+        # --eval_every was on, else the guard's best-training-tau snapshot
+        ship = best_eval_params if best_eval_params is not None else best_params
+        if ship is not None:
             with torch.no_grad():
-                for p, bp in zip(engine.model.parameters(), best_params):
+                for p, bp in zip(engine.model.parameters(), ship):
                     p.copy_(bp)
         ckpt_io.save(engine.model, os.path.join(actor_dir, "cp-9999.ckpt"))
         with open(os.path.join(args.model_root,

@@ -58,3 +58,35 @@ def test_demo_main(tmp_path):
     assert any("flow_routes" in f for f in figs)
     assert any("flow_packets" in f for f in figs)
     assert os.listdir(tmp_path / "out")
+
+
+def test_train_batched_eval_selection(tmp_path):
+    """--eval_every: held-out eval records land in the history and the
+    shipped cp-9999 equals the best-by-eval-tau parameters."""
+    import numpy as np
+    import torch
+    from multihop_offload_amd.harness import train_batched
+    from multihop_offload_amd.utils import checkpoint as ckpt_io
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+
+    history = train_batched.main([
+        "--steps", "6", "--batch", "8", "--sizes", "20", "--distinct", "4",
+        "--workers", "0", "--seed", "5", "--device", "cpu",
+        "--eval_every", "3", "--eval_rounds", "1", "--guard_every", "0",
+        "--save_every", "100", "--log_every", "100",
+        "--model_root", str(tmp_path), "--training_set", "EVT"])
+    evals = [h for h in history if "eval_tau" in h]
+    assert len(evals) == 2 and all(np.isfinite(h["eval_tau"]) for h in evals)
+    best = min(h["eval_tau"] for h in evals)
+    assert evals[-1]["best_eval_tau"] == best
+
+    m = ChebConvStack(K=2, dtype=torch.float64, seed=0)
+    ckpt_io.load(m, str(tmp_path / "model_ChebConv_EVT_a5_c5_ACO_agent"
+                        / "cp-9999.ckpt"))
+    # reproduce the eval at the shipped parameters: matches best_eval_tau
+    cases = train_batched.build_training_cases(20, 8, 4, 1000, 5 + 17 * 20,
+                                               workers=0)
+    from multihop_offload_amd.engine import EpisodeEngine
+    eng = EpisodeEngine(cases, m, device="cpu", dtype=torch.float64)
+    tau = train_batched.evaluate_policy([eng], [0.15], 12345, rounds=1)
+    assert np.isclose(tau, best, rtol=1e-6), (tau, best)
