@@ -338,6 +338,35 @@ class EpisodeEngine:
             dl=torch.where(mask, torch.full_like(z, 1.0), z),
         )
 
+    def pack_jobs(self, instances) -> JobBatch:
+        """Pack a list of per-case :class:`JobInstance` (length B, each with
+        ``num_jobs <= Jmax``) into a padded device JobBatch — the bridge
+        from the host-side samplers/harnesses into the engine."""
+        B, J = self.B, self.Jmax
+        assert len(instances) == B
+        src = np.zeros((B, J), dtype=np.int64)
+        mask = np.zeros((B, J), dtype=bool)
+        rates = np.zeros((B, J))
+        ul = np.zeros((B, J))
+        dl = np.zeros((B, J))
+        for b, jobs in enumerate(instances):
+            k = jobs.num_jobs
+            assert k <= J, (k, J)
+            src[b, :k] = jobs.sources
+            src[b, k:] = jobs.sources[0] if k else 0
+            mask[b, :k] = True
+            rates[b, :k] = jobs.rates
+            ul[b, :k] = jobs.ul
+            dl[b, :k] = jobs.dl
+        dev, dt = self.device, self.dtype
+        return JobBatch(
+            sources=torch.as_tensor(src, device=dev),
+            mask=torch.as_tensor(mask, device=dev),
+            rates=torch.as_tensor(rates, dtype=dt, device=dev),
+            ul=torch.as_tensor(ul, dtype=dt, device=dev),
+            dl=torch.as_tensor(dl, dtype=dt, device=dev),
+        )
+
     # ------------------------------------------------------- actor forward
     def actor_forward(self, jobs: JobBatch):
         """features → ChebConv → λ → fixed point → delays → (B,N,N) delay

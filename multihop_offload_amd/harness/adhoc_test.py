@@ -48,6 +48,8 @@ def main(argv=None):
         filepath = os.path.join(args.datapath, names[fid])
         g = common.load_case(filepath, args.T, rng)
         env = AdhocCloudEnv(g)
+        runner = (common.EngineRunner(agent, g, seed=args.seed or 0)
+                  if args.engine else None)
         t_case = time.time()
         for ni in range(args.instances):
             jobs = JobInstance.sample(g.mobile_nodes, args.arrival_scale,
@@ -56,7 +58,9 @@ def main(argv=None):
             for method in ["baseline", "local", "GNN"]:
                 env.set_jobs(jobs)
                 t0 = time.time()
-                delay_emp = common.run_method(method, agent, env, 0.0, rng)
+                delay_emp = (runner.run_method(method, jobs, 0.0)
+                             if runner is not None else
+                             common.run_method(method, agent, env, 0.0, rng))
                 runtime = time.time() - t0
                 delay_dict[method] = delay_emp
                 rows.append({

@@ -52,6 +52,8 @@ def main(argv=None):
             filepath = os.path.join(args.datapath, names[fid])
             g = common.load_case(filepath, args.T, rng)
             env = AdhocCloudEnv(g)
+            runner = (common.EngineRunner(agent, g, seed=args.seed or 0)
+                      if args.engine else None)
             for ni in range(args.instances):
                 jobs = JobInstance.sample(g.mobile_nodes, args.arrival_scale,
                                           rng or np.random)
@@ -59,8 +61,10 @@ def main(argv=None):
                 for method in ["baseline", "local", "GNN", "GNN-test"]:
                     env.set_jobs(jobs)
                     t0 = time.time()
-                    delay_emp = common.run_method(method, agent, env,
-                                                  explore, rng)
+                    delay_emp = (runner.run_method(method, jobs, explore)
+                                 if runner is not None else
+                                 common.run_method(method, agent, env,
+                                                   explore, rng))
                     runtime = time.time() - t0
                     delay_dict[method] = delay_emp
                     rows.append({

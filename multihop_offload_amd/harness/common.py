@@ -24,6 +24,10 @@ def build_parser(default_datapath="../data_100") -> argparse.ArgumentParser:
     add("--out", type=str, default="out", help="output data path")
     add("--T", type=int, default=1000)
     add("--prob", action="store_true", help="probabilistic decision")
+    add("--engine", action="store_true",
+        help="run all method branches through the batched device engine "
+             "(B=1 per case) instead of the numpy oracle — the fused HIP "
+             "path on GPU; identical numbers at explore=0")
     add("--training_set", type=str, default="BAm2")
     add("--learning_rate", type=float, default=0.0001)
     add("--learning_decay", type=float, default=1.0)
@@ -108,3 +112,50 @@ def list_cases(datapath: str, limit: int = 0):
     if limit:
         names = names[:limit]
     return names
+
+
+class EngineRunner:
+    """Engine-backed method execution for the compat harnesses
+    (``--engine``): one B=1 :class:`EpisodeEngine` per case visit, so the
+    reference workflow runs on the fused HIP path on GPU instead of the
+    numpy oracle.  Numerically identical to ``run_method`` at explore=0
+    (test_engine's B=1 oracle equivalences); with explore>0 the decision
+    RNG is the device generator rather than numpy, so individual draws
+    differ while the statistics match.  The GNN branch memorises the
+    per-instance gradient set exactly like ``ACOAgent.forward_backward``
+    (B=1: the summed engine gradient IS the instance gradient)."""
+
+    def __init__(self, agent: ACOAgent, g, seed: int = 0):
+        import torch
+        from ..engine import EpisodeEngine
+        self.agent = agent
+        self.engine = EpisodeEngine([g], agent.model,
+                                    device=str(agent.device),
+                                    dtype=agent.dtype)
+        self.gen = torch.Generator(device=str(agent.device))
+        self.gen.manual_seed(seed * 9973 + g.seed)
+
+    def run_method(self, method: str, jobs, explore: float = 0.0):
+        import torch
+        eng = self.engine
+        jb = eng.pack_jobs([jobs])
+        if method == "baseline":
+            res = eng.baseline_episode(jb)
+        elif method == "local":
+            res = eng.local_episode(jb)
+        elif method == "GNN":
+            for p in eng.model.parameters():
+                p.grad = None
+            res = eng.gnn_episode(jb, explore=explore, gen=self.gen,
+                                  train=True, prob=self.agent.cfg.prob)
+            grads = [p.grad.detach().clone() if p.grad is not None
+                     else torch.zeros_like(p)
+                     for p in eng.model.parameters()]
+            self.agent.memorize(grads, float(res.loss_fn),
+                                float(res.loss_mse))
+        elif method == "GNN-test":
+            res = eng.gnn_episode(jb, train=False)
+        else:
+            raise ValueError(method)
+        k = jobs.num_jobs
+        return res.delay_emp[0, :k].detach().cpu().numpy()

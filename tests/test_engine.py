@@ -26,28 +26,7 @@ def _case(seed=7, n=20):
 
 def _jobbatch_from(engine, instances):
     """Pack per-case JobInstance lists into a padded JobBatch."""
-    B, J = engine.B, engine.Jmax
-    src = np.zeros((B, J), dtype=np.int64)
-    mask = np.zeros((B, J), dtype=bool)
-    rates = np.zeros((B, J))
-    ul = np.zeros((B, J))
-    dl = np.zeros((B, J))
-    for b, jobs in enumerate(instances):
-        k = jobs.num_jobs
-        src[b, :k] = jobs.sources
-        src[b, k:] = jobs.sources[0] if k else 0
-        mask[b, :k] = True
-        rates[b, :k] = jobs.rates
-        ul[b, :k] = jobs.ul
-        dl[b, :k] = jobs.dl
-    dev, dt = engine.device, engine.dtype
-    return JobBatch(
-        sources=torch.as_tensor(src, device=dev),
-        mask=torch.as_tensor(mask, device=dev),
-        rates=torch.as_tensor(rates, dtype=dt, device=dev),
-        ul=torch.as_tensor(ul, dtype=dt, device=dev),
-        dl=torch.as_tensor(dl, dtype=dt, device=dev),
-    )
+    return engine.pack_jobs(instances)
 
 
 def _wake(model):
