@@ -353,7 +353,10 @@ class EpisodeEngine:
             k = jobs.num_jobs
             assert k <= J, (k, J)
             src[b, :k] = jobs.sources
-            src[b, k:] = jobs.sources[0] if k else 0
+            # pad with a node that has a virtual self-loop edge: a relay
+            # (node 0 can be one) has none and would index -1 downstream
+            src[b, k:] = (jobs.sources[0] if k
+                          else self.cases[b].ext.comp_nodes[0])
             mask[b, :k] = True
             rates[b, :k] = jobs.rates
             ul[b, :k] = jobs.ul

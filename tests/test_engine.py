@@ -301,3 +301,53 @@ def test_walk_overflow_guard():
     rl, nhop = engine.route_walk(jobs, far, sp)
     # capped at walk_cap hops; unreached jobs simply stop
     assert int(nhop.max()) <= 1
+
+
+@pytest.mark.parametrize("seed", range(8))
+def test_engine_oracle_sweep_random_topologies(seed):
+    """Randomized hardening sweep: for random topologies (mixed families),
+    random role assignments from the datagen distributions, and random job
+    draws, the batched engine reproduces the oracle's GNN-episode delays
+    and parameter gradients exactly (fp64, explore=0)."""
+    from multihop_offload_amd.datagen import generate_case
+    from multihop_offload_amd.harness.common import run_method
+    from multihop_offload_amd.agent import ACOAgent, AgentConfig
+
+    rng = np.random.RandomState(1000 + seed)
+    n = int(rng.choice([12, 16, 20, 25]))
+    gtype = ["ba", "er", "ws"][seed % 3]
+    case = generate_case(n, 3000 + seed, gtype, rng=rng)[0]
+    g = CaseGraph(n, t_max=1000, seed=3000 + seed, gtype=gtype,
+                  adj=np.asarray(case["adj"].todense()), pos=case["pos_c"])
+    for nidx in range(n):
+        role, bw = case["nodes_info"][nidx, 0], float(
+            case["nodes_info"][nidx, 1])
+        if role == 2:
+            g.add_relay(nidx)
+        elif role == 1:
+            g.add_server(nidx, bw)
+        else:
+            g.set_mobile_bw(nidx, bw)
+    g.links_init(case["link_rate"], rng=rng)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, rng)
+
+    agent = ACOAgent(AgentConfig(seed=seed), 10)
+    _wake(agent.model)
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    d_oracle = run_method("GNN", agent, env, 0.0, np.random.RandomState(0))
+
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    model2 = ChebConvStack(K=2, dtype=torch.float64, seed=seed)
+    _wake(model2)
+    engine = EpisodeEngine([g], model2, device="cpu", dtype=torch.float64)
+    for p in model2.parameters():
+        p.grad = None
+    res = engine.gnn_episode(engine.pack_jobs([jobs]), train=True)
+    np.testing.assert_allclose(
+        res.delay_emp[0, :jobs.num_jobs].detach().numpy(), d_oracle,
+        rtol=1e-9)
+    grads_oracle = agent.memory[-1][0]
+    for p, go in zip(model2.parameters(), grads_oracle):
+        assert torch.allclose(p.grad, go, atol=1e-9), \
+            (p.shape, (p.grad - go).abs().max())
