@@ -225,10 +225,10 @@ def main(argv=None):
                 best_eval_tau = eval_tau
                 best_eval_params = [p.detach().clone()
                                     for p in engine.model.parameters()]
+            rec = {"step": step, "eval_tau": eval_tau,
+                   "best_eval_tau": best_eval_tau}
+            history.append(rec)        # every rank: values are all-reduced
             if rank == 0:
-                rec = {"step": step, "eval_tau": eval_tau,
-                       "best_eval_tau": best_eval_tau}
-                history.append(rec)
                 print(json.dumps(rec), flush=True)
         if args.guard_every and step % args.guard_every == 0:
             tau_now = _dist_mean(float(torch.nanmean(res.tau)),

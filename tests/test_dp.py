@@ -66,3 +66,49 @@ def test_engine_rank_sharded_bench_step():
     res = engine.gnn_episode(jobs, train=True, gen=gen)
     assert torch.isfinite(res.tau).all()
     assert res.loss_fn is not None and np.isfinite(float(res.loss_fn))
+
+
+def _trainer_worker(rank, world, port, tmpdir, q):
+    import torch.distributed as dist
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    try:
+        from multihop_offload_amd.harness import train_batched
+        history = train_batched.main([
+            "--steps", "4", "--batch", "8", "--sizes", "20",
+            "--distinct", "4", "--workers", "0", "--seed", "5",
+            "--device", "cpu", "--eval_every", "2", "--eval_rounds", "1",
+            "--guard_every", "2", "--save_every", "100",
+            "--log_every", "100", "--model_root", tmpdir,
+            "--training_set", "DP2"])
+        evals = [h["eval_tau"] for h in history if "eval_tau" in h]
+        q.put((rank, evals))
+        dist.destroy_process_group()
+    except Exception as e:         # surface failures through the queue
+        q.put((rank, f"ERR {type(e).__name__}: {e}"))
+
+
+def test_train_batched_gloo_ws2(tmp_path):
+    """Full trainer main() at world=2 over gloo: eval/guard decisions are
+    all-reduced (rank-coherent) and the run completes on both ranks."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    ps = [ctx.Process(target=_trainer_worker,
+                      args=(r, 2, 29653, str(tmp_path), q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(300)
+    results = {}
+    while not q.empty():
+        r, val = q.get()
+        results[r] = val
+    assert isinstance(results.get(0), list), results
+    assert isinstance(results.get(1), list), results
+    # eval taus are all-reduced: both ranks saw identical values
+    assert len(results[0]) == 2 and results[0] == results[1], results
+    assert (tmp_path / "model_ChebConv_DP2_a5_c5_ACO_agent"
+            / "cp-9999.ckpt.npz").exists()
