@@ -271,6 +271,41 @@ class CaseGraph:
             g._ext = e
         return g
 
+    def pad_to(self, n_target: int) -> "CaseGraph":
+        """Inert node padding for mixed-size batching in ONE engine: a new
+        case with ``n_target - num_nodes`` additional isolated relay nodes
+        (no links, no virtual edges, bw 0).  Behaviour-invariant end to
+        end: the link set, canonical ordering, conflict CSR and extended
+        graph are identical to the original's; padded nodes are
+        unreachable (inf APSP rows), never sampled as job sources (not
+        mobile), never chosen as servers, and contribute no features.
+        This lets batches mix graph sizes without per-size engine buckets
+        and without any kernel change."""
+        assert n_target >= self.num_nodes
+        if n_target == self.num_nodes:
+            return self
+        N0 = self.num_nodes
+        adj = np.zeros((n_target, n_target), dtype=np.int8)
+        adj[:N0, :N0] = self.adj
+        pos = None
+        if self.pos is not None:
+            pos = np.vstack([np.asarray(self.pos, dtype=np.float64),
+                             np.zeros((n_target - N0, 2))])
+        g = CaseGraph(n_target, t_max=self.T, seed=self.seed, m=self.m,
+                      gtype=self.gtype, adj=adj, pos=pos,
+                      cf_radius=self.cf_radius)
+        for v in range(N0):
+            if self.roles[v] == 2:
+                g.add_relay(v)
+            elif self.roles[v] == 1:
+                g.add_server(v, self.proc_bws[v])
+            else:
+                g.set_mobile_bw(v, self.proc_bws[v])
+        for v in range(N0, n_target):
+            g.add_relay(v)
+        g.link_rates = self.link_rates.copy()
+        return g
+
     # -- conflict radius augmentation (offloading_v3.py:193-224) --------------
     def _add_conflict_relations(self, conf_sets):
         if self.pos is None:

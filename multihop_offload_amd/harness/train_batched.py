@@ -142,6 +142,11 @@ def main(argv=None):
                          "back to the best parameters and cut lr 3x")
     ap.add_argument("--init_scale", type=float, default=0.01,
                     help="shrink initial weights (wakes the output ReLU)")
+    ap.add_argument("--pad_mixed", action="store_true",
+                    help="mixed --sizes in ONE engine (smaller cases "
+                         "padded with inert relay nodes to the max size): "
+                         "every step trains on the full mixed batch "
+                         "instead of round-robining per-size engines")
     ap.add_argument("--eval_every", type=int, default=0,
                     help="held-out greedy evaluation period (0 disables); "
                          "the final checkpoint is then the best-by-eval-tau "
@@ -164,13 +169,23 @@ def main(argv=None):
 
     sizes = ([int(s) for s in args.sizes.split(",")] if args.sizes
              else [args.nodes])
-    engines = []
-    for n in sizes:
-        cases = build_training_cases(
-            n, max(args.batch // len(sizes), 8), args.distinct, args.T,
-            args.seed + 1000 * rank + 17 * n, workers=args.workers)
-        engines.append(EpisodeEngine(cases, model, device=device,
-                                     dtype=dtype))
+    per_size = max(args.batch // len(sizes), 8)
+    if args.pad_mixed and len(sizes) > 1:
+        n_max = max(sizes)
+        cases = []
+        for n in sizes:
+            cases += [c.pad_to(n_max) for c in build_training_cases(
+                n, per_size, args.distinct, args.T,
+                args.seed + 1000 * rank + 17 * n, workers=args.workers)]
+        engines = [EpisodeEngine(cases, model, device=device, dtype=dtype)]
+    else:
+        engines = []
+        for n in sizes:
+            cases = build_training_cases(
+                n, per_size, args.distinct, args.T,
+                args.seed + 1000 * rank + 17 * n, workers=args.workers)
+            engines.append(EpisodeEngine(cases, model, device=device,
+                                         dtype=dtype))
     engine = engines[0]
     dp.broadcast_params(engine.model)
     opt = torch.optim.Adam(engine.model.parameters(),

@@ -351,3 +351,46 @@ def test_engine_oracle_sweep_random_topologies(seed):
     for p, go in zip(model2.parameters(), grads_oracle):
         assert torch.allclose(p.grad, go, atol=1e-9), \
             (p.shape, (p.grad - go).abs().max())
+
+
+def test_padded_case_is_behaviour_invariant():
+    """pad_to: a case padded with inert relay nodes produces identical
+    episode delays AND parameter gradients; a mixed-size batch in ONE
+    engine equals the per-size engines."""
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+
+    g20, g25 = _case(seed=3, n=20), _case(seed=9, n=25)
+    rng = np.random.RandomState(0)
+    j20 = JobInstance.sample(g20.mobile_nodes, 0.15, rng)
+    j25 = JobInstance.sample(g25.mobile_nodes, 0.15, rng)
+
+    def run(cases, instances):
+        model = ChebConvStack(K=2, dtype=torch.float64, seed=11)
+        _wake(model)
+        eng = EpisodeEngine(cases, model, device="cpu",
+                            dtype=torch.float64)
+        for p in model.parameters():
+            p.grad = None
+        res = eng.gnn_episode(eng.pack_jobs(instances), train=True)
+        grads = [p.grad.clone() for p in model.parameters()]
+        return eng, res, grads
+
+    # B=1: padded == original exactly
+    _, r_orig, g_orig = run([g20], [j20])
+    _, r_pad, g_pad = run([g20.pad_to(25)], [j20])
+    np.testing.assert_allclose(r_pad.delay_emp[0, :j20.num_jobs],
+                               r_orig.delay_emp[0, :j20.num_jobs],
+                               rtol=1e-12)
+    for a, b in zip(g_pad, g_orig):
+        assert torch.allclose(a, b, atol=1e-12)
+
+    # mixed-size batch in one engine == sum of the two singles
+    _, r25, g25g = run([g25], [j25])
+    _, r_mix, g_mix = run([g20.pad_to(25), g25], [j20, j25])
+    np.testing.assert_allclose(r_mix.delay_emp[0, :j20.num_jobs],
+                               r_orig.delay_emp[0, :j20.num_jobs],
+                               rtol=1e-12)
+    np.testing.assert_allclose(r_mix.delay_emp[1, :j25.num_jobs],
+                               r25.delay_emp[0, :j25.num_jobs], rtol=1e-12)
+    for m, a, b in zip(g_mix, g_orig, g25g):
+        assert torch.allclose(m, a + b, atol=1e-9), (m - a - b).abs().max()

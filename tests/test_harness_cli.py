@@ -169,3 +169,19 @@ def test_evaluate_cli_with_shipped_checkpoint(tmp_path):
     assert s["GNN"]["tau"] > 0 and s["GNN"]["latency_ratio"] < 2.0
     assert s["GNN"]["congest_ratio"] <= s["baseline"]["congest_ratio"] + 1e-9
     assert blob["per_size"]
+
+
+def test_train_batched_pad_mixed(tmp_path):
+    """--pad_mixed trains mixed sizes in a single padded engine."""
+    import numpy as np
+    from multihop_offload_amd.harness import train_batched
+    history = train_batched.main([
+        "--steps", "4", "--batch", "16", "--sizes", "16,20",
+        "--distinct", "4", "--workers", "0", "--seed", "6",
+        "--device", "cpu", "--pad_mixed", "--eval_every", "2",
+        "--eval_rounds", "1", "--guard_every", "0", "--save_every", "100",
+        "--log_every", "2", "--model_root", str(tmp_path),
+        "--training_set", "PMX"])
+    assert any("eval_tau" in h and np.isfinite(h["eval_tau"])
+               for h in history)
+    assert any("tau" in h and np.isfinite(h["tau"]) for h in history)
