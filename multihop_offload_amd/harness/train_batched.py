@@ -221,6 +221,9 @@ def main(argv=None):
                     p.grad /= engine.B
         reducer(average=True)
         with torch.no_grad():
+            pre_norms = [float(p.grad.norm())
+                         for p in engine.model.parameters()
+                         if p.grad is not None]
             for p in engine.model.parameters():
                 if p.grad is not None:
                     n = p.grad.norm().clamp(min=1e-12)
@@ -276,7 +279,14 @@ def main(argv=None):
             gnorm = float(sum((p.grad ** 2).sum()
                               for p in engine.model.parameters()
                               if p.grad is not None) ** 0.5)
+            # pre-clip norms are the stability signal: the post-clip norm
+            # is pinned at sqrt(#tensors) whenever every tensor saturates
+            # the per-tensor clip (see ROUND2.md stability probe)
             rec = {"step": step, "tau": tau,
+                   "grad_norm_preclip_max": max(pre_norms) if pre_norms
+                   else 0.0,
+                   "grad_norm_preclip_sum": float(np.hypot.reduce(
+                       pre_norms)) if pre_norms else 0.0,
                    "congest_ratio": congest / max(njobs, 1),
                    "loss_fn": float(res.loss_fn), "loss_mse": float(res.loss_mse),
                    "grad_norm": gnorm,
