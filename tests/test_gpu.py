@@ -356,3 +356,84 @@ def test_per_sample_gradients_match_oracle():
             a = got.cpu().numpy()
             w = want.cpu().numpy()
             assert np.abs(a - w).max() / max(np.abs(w).max(), 1e-6) < 1e-2
+
+
+@needs_gpu
+def test_padded_mixed_batch_gpu():
+    """pad_to invariance on the HIP path: a case padded with inert relay
+    nodes yields the same delays and gradients as the original, and a
+    mixed-size batch in one engine sums the per-case gradients."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.graphs import JobInstance
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from tests.test_engine import _case, _wake
+
+    g20, g25 = _case(seed=3, n=20), _case(seed=9, n=25)
+    rng = np.random.RandomState(0)
+    j20 = JobInstance.sample(g20.mobile_nodes, 0.15, rng)
+    j25 = JobInstance.sample(g25.mobile_nodes, 0.15, rng)
+
+    def run(cases, instances):
+        model = ChebConvStack(K=2, dtype=torch.float32, seed=11)
+        _wake(model)
+        eng = EpisodeEngine(cases, model, device="cuda",
+                            dtype=torch.float32)
+        assert eng.use_hip
+        for p in model.parameters():
+            p.grad = None
+        res = eng.gnn_episode(eng.pack_jobs(instances), train=True)
+        torch.cuda.synchronize()
+        grads = [p.grad.clone() for p in model.parameters()]
+        return res, grads
+
+    r_orig, g_orig = run([g20], [j20])
+    r_pad, g_pad = run([g20.pad_to(25)], [j20])
+    torch.testing.assert_close(r_pad.delay_emp[0, :j20.num_jobs],
+                               r_orig.delay_emp[0, :j20.num_jobs],
+                               rtol=1e-5, atol=1e-4, equal_nan=True)
+    for a, b in zip(g_pad, g_orig):
+        torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-5)
+
+    r25, g25g = run([g25], [j25])
+    r_mix, g_mix = run([g20.pad_to(25), g25], [j20, j25])
+    torch.testing.assert_close(r_mix.delay_emp[0, :j20.num_jobs],
+                               r_orig.delay_emp[0, :j20.num_jobs],
+                               rtol=1e-5, atol=1e-4, equal_nan=True)
+    torch.testing.assert_close(r_mix.delay_emp[1, :j25.num_jobs],
+                               r25.delay_emp[0, :j25.num_jobs],
+                               rtol=1e-5, atol=1e-4, equal_nan=True)
+    for m, a, b in zip(g_mix, g_orig, g25g):
+        torch.testing.assert_close(m, a + b, rtol=1e-3, atol=1e-4)
+
+
+@needs_gpu
+def test_engine_runner_gpu_matches_cpu_oracle():
+    """The --engine harness path on the HIP kernels agrees with the CPU
+    fp64 oracle at explore=0 (fp32 tolerances)."""
+    from multihop_offload_amd import ACOAgent
+    from multihop_offload_amd.agent import AgentConfig
+    from multihop_offload_amd.env import AdhocCloudEnv
+    from multihop_offload_amd.harness import common
+    from multihop_offload_amd.graphs import JobInstance
+    from tests.test_engine import _case, _wake
+
+    g = _case(seed=13, n=20)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, np.random.RandomState(3))
+
+    a_cpu = ACOAgent(AgentConfig(seed=4, dtype="float64"), 10)
+    _wake(a_cpu.model)
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    d_oracle = common.run_method("GNN", a_cpu, env, 0.0,
+                                 np.random.RandomState(0))
+
+    a_gpu = ACOAgent(AgentConfig(seed=4, device="cuda",
+                                 dtype="float32"), 10)
+    with torch.no_grad():
+        for pc, pg in zip(a_cpu.model.parameters(),
+                          a_gpu.model.parameters()):
+            pg.copy_(pc.to(torch.float32))
+    runner = common.EngineRunner(a_gpu, g, seed=4)
+    assert runner.engine.use_hip
+    d_engine = runner.run_method("GNN", jobs, 0.0)
+    np.testing.assert_allclose(d_engine, d_oracle, rtol=1e-3, atol=1e-2)
