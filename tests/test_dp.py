@@ -112,3 +112,30 @@ def test_train_batched_gloo_ws2(tmp_path):
     assert len(results[0]) == 2 and results[0] == results[1], results
     assert (tmp_path / "model_ChebConv_DP2_a5_c5_ACO_agent"
             / "cp-9999.ckpt.npz").exists()
+
+
+def test_bench_distributed_ws2_cpu(tmp_path):
+    """bench.py under torch.distributed.run, world=2, gloo on CPU: the
+    driver's multi-GPU launch shape.  Rank 0 must print one valid JSON
+    line with the contract fields and n_gpus=2."""
+    import json
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    env.pop("RANK", None); env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29671", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--batch", "16", "--nodes", "20",
+         "--distinct", "4", "--device", "cpu"],
+        capture_output=True, text=True, timeout=420, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-2000:]
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["steps"] == 2
+    assert rec["value"] > 0 and rec["scaling"] == "weak"
+    assert {"metric", "unit", "ms_per_step", "higher_is_better",
+            "dtype", "data", "config"} <= set(rec)
