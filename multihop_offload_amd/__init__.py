@@ -19,3 +19,4 @@ __version__ = "0.1.0"
 from .graphs import CaseGraph, JobInstance  # noqa: F401
 from .env import AdhocCloudEnv, AdhocCloud  # noqa: F401
 from .agent import ACOAgent                 # noqa: F401
+from .engine import EpisodeEngine, JobBatch, EpisodeResult  # noqa: F401
