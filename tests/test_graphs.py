@@ -143,3 +143,92 @@ def test_mobility_random_walk_and_topology_update():
             assert m[l] == old_lm[u, v]
         else:
             assert m[l] == -1
+
+
+def test_cf_radius_conflict_augmentation():
+    """cf_radius > 0.5 adds interference conflicts between links whose
+    endpoints lie within cf_radius x median link distance — checked
+    against a literal networkx transcription of the reference rule
+    (offloading_v3.py:193-224)."""
+    import networkx as nx
+    from scipy.spatial import distance_matrix as dmat
+
+    g = CaseGraph(30, seed=5, gtype="poisson", m=4, cf_radius=1.5)
+    base = CaseGraph(30, seed=5, gtype="poisson", m=4)   # same topology
+    assert np.array_equal(g.edges, base.edges)
+    pos = np.asarray(g.pos)
+
+    # literal transcription
+    d = dmat(pos, pos)
+    link_dist = d[g.edges[:, 0], g.edges[:, 1]]
+    intf = 1.5 * np.nanmedian(link_dist)
+    gi = nx.Graph()
+    gi.add_nodes_from(range(g.num_links))
+    # base conflicts: shared endpoint
+    for a in range(g.num_links):
+        for b in range(a + 1, g.num_links):
+            if set(g.edges[a]) & set(g.edges[b]):
+                gi.add_edge(a, b)
+    for l, (u, v) in enumerate(g.edges):
+        near = np.union1d(np.nonzero(d[u] < intf)[0],
+                          np.nonzero(d[v] < intf)[0])
+        for w in near:
+            for x in np.nonzero(g.adj[w])[0]:
+                ll = g.link_matrix[w, x]
+                if ll >= 0 and ll != l:
+                    gi.add_edge(l, int(ll))
+
+    for l in range(g.num_links):
+        want = sorted(gi.neighbors(l))
+        lo, hi = g.conf_indptr[l], g.conf_indptr[l + 1]
+        got = sorted(g.conf_indices[lo:hi])
+        assert got == want, (l, got, want)
+    # augmentation strictly grows the conflict degree somewhere
+    assert g.cf_degs.sum() > base.cf_degs.sum()
+
+
+def test_cf_radius_engine_matches_oracle():
+    """A cf_radius-augmented case runs through the batched engine with the
+    same delays/gradients as the oracle (the conflict CSR feeds the fixed
+    point everywhere)."""
+    import torch
+    from multihop_offload_amd.agent import ACOAgent, AgentConfig
+    from multihop_offload_amd.env import AdhocCloudEnv
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.harness.common import run_method
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+
+    rng = np.random.RandomState(7)
+    g = CaseGraph(30, seed=5, gtype="poisson", m=4, cf_radius=1.5)
+    g.links_init(50.0, rng=rng)
+    g.add_relay(0)
+    for s in (2, 3, 4):
+        g.add_server(s, 300.0)
+    for v in range(5, 30):
+        if g.roles[v] == 0:
+            g.set_mobile_bw(v, 10.0)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, rng)
+
+    agent = ACOAgent(AgentConfig(seed=2), 10)
+    for layer in agent.model.layers:
+        with torch.no_grad():
+            layer.weight.mul_(0.01)
+    with torch.no_grad():
+        agent.model.layers[-1].bias.fill_(0.5)
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    d_oracle = run_method("GNN", agent, env, 0.0, np.random.RandomState(0))
+
+    model2 = ChebConvStack(K=2, dtype=torch.float64, seed=2)
+    with torch.no_grad():
+        for layer in model2.layers:
+            layer.weight.mul_(0.01)
+        model2.layers[-1].bias.fill_(0.5)
+    eng = EpisodeEngine([g], model2, device="cpu", dtype=torch.float64)
+    for p in model2.parameters():
+        p.grad = None
+    res = eng.gnn_episode(eng.pack_jobs([jobs]), train=True)
+    np.testing.assert_allclose(res.delay_emp[0, :jobs.num_jobs], d_oracle,
+                               rtol=1e-9)
+    for p, go in zip(model2.parameters(), agent.memory[-1][0]):
+        assert torch.allclose(p.grad, go, atol=1e-9)
