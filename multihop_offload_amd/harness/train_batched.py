@@ -154,6 +154,9 @@ def main(argv=None):
                          "training tau the guard tracks")
     ap.add_argument("--eval_rounds", type=int, default=4)
     ap.add_argument("--eval_seed", type=int, default=12345)
+    ap.add_argument("--resume", action="store_true",
+                    help="load the latest checkpoint from the model dir "
+                         "before training (reference resume protocol)")
     args = ap.parse_args(argv)
 
     rank, world = dp.init_from_env()
@@ -161,7 +164,19 @@ def main(argv=None):
     dtype = torch.float32 if device.startswith("cuda") else torch.float64
 
     model = ChebConvStack(K=args.K, dtype=dtype, seed=args.seed)
-    if args.init_scale != 1.0:
+    resumed = False
+    if args.resume:
+        from ..utils.checkpoint import latest_checkpoint
+        latest = latest_checkpoint(model_dir(args.model_root,
+                                             args.training_set))
+        if latest:
+            ckpt_io.load(model, latest)
+            resumed = True
+            if rank == 0:
+                print(f"resumed from {latest}", flush=True)
+        elif rank == 0:
+            print("no checkpoint to resume — fresh init", flush=True)
+    if args.init_scale != 1.0 and not resumed:
         with torch.no_grad():
             for layer in model.layers:
                 layer.weight.mul_(args.init_scale)

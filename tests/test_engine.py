@@ -394,3 +394,48 @@ def test_padded_case_is_behaviour_invariant():
                                r25.delay_emp[0, :j25.num_jobs], rtol=1e-12)
     for m, a, b in zip(g_mix, g_orig, g25g):
         assert torch.allclose(m, a + b, atol=1e-9), (m - a - b).abs().max()
+
+
+def test_prob_mode_softmax_statistics():
+    """prob=True samples destinations with the reference's (non-negated,
+    high-cost-preferring) softmax over costs: empirical frequencies from
+    the engine match the oracle softmax probabilities."""
+    from multihop_offload_amd.env import AdhocCloudEnv, apsp, softmax
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+
+    g = _case(seed=7, n=20)
+    rng = np.random.RandomState(1)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, rng)
+    model = ChebConvStack(K=2, dtype=torch.float64, seed=3)
+    _wake(model)
+    eng = EpisodeEngine([g], model, device="cpu", dtype=torch.float64)
+    jb = eng.pack_jobs([jobs])
+
+    with torch.no_grad():
+        dm, link_delay, _ = eng.actor_forward(jb)
+    sp = eng.apsp(dm)
+    uds = torch.diagonal(dm, dim1=1, dim2=2)
+
+    # oracle probabilities for job 0 from the same GNN delays
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    sp_np = apsp(g, link_delay[0, :g.num_links].numpy())
+    np.fill_diagonal(sp_np, uds[0].numpy())
+    servers = np.asarray(g.servers)
+    spz = sp_np.copy(); np.fill_diagonal(spz, 0)
+    s0 = int(jobs.sources[0])
+    ul_d = np.maximum(spz[s0, servers] * jobs.ul[0], g.sp_hop[s0, servers])
+    dl_d = np.maximum(spz[servers, s0] * jobs.dl[0], g.sp_hop[servers, s0])
+    pr_d = np.maximum(sp_np[servers, servers] * jobs.ul[0], 1.0)
+    local = sp_np[s0, s0] * jobs.ul[0]
+    p_want = softmax(np.concatenate([ul_d + dl_d + pr_d, [local]]))
+
+    gen = torch.Generator().manual_seed(0)
+    counts = np.zeros(len(servers) + 1)
+    trials = 4000
+    for _ in range(trials):
+        dst, _ = eng.offload_decide(jb, sp, uds, 0.0, gen, prob=True)
+        d0 = int(dst[0, 0])
+        idx = (list(servers).index(d0) if d0 in servers else len(servers))
+        counts[idx] += 1
+    np.testing.assert_allclose(counts / trials, p_want, atol=0.03)

@@ -185,3 +185,34 @@ def test_train_batched_pad_mixed(tmp_path):
     assert any("eval_tau" in h and np.isfinite(h["eval_tau"])
                for h in history)
     assert any("tau" in h and np.isfinite(h["tau"]) for h in history)
+
+
+def test_train_batched_resume(tmp_path):
+    """--resume continues from the latest checkpoint (reference resume
+    protocol): the resumed run starts from the saved parameters."""
+    import torch
+    from multihop_offload_amd.harness import train_batched
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.utils import checkpoint as ckpt_io
+    from multihop_offload_amd.utils.checkpoint import (latest_checkpoint,
+                                                       model_dir)
+    common = ["--batch", "8", "--sizes", "20", "--distinct", "4",
+              "--workers", "0", "--seed", "5", "--device", "cpu",
+              "--guard_every", "0", "--save_every", "100",
+              "--log_every", "100", "--model_root", str(tmp_path),
+              "--training_set", "RSM"]
+    train_batched.main(["--steps", "3"] + common)
+    d = model_dir(str(tmp_path), "RSM")
+    saved = latest_checkpoint(d)
+    assert saved
+    m_saved = ChebConvStack(K=2, dtype=torch.float64, seed=99)
+    ckpt_io.load(m_saved, saved)
+
+    # second run with --resume must start from (not re-randomize) those
+    # parameters; after 1 step with lr 0 nothing changes
+    train_batched.main(["--steps", "1", "--learning_rate", "0.0",
+                        "--resume"] + common)
+    m_after = ChebConvStack(K=2, dtype=torch.float64, seed=123)
+    ckpt_io.load(m_after, latest_checkpoint(d))
+    for a, b in zip(m_after.parameters(), m_saved.parameters()):
+        assert torch.allclose(a, b, atol=1e-12)
