@@ -263,7 +263,6 @@ class ACOAgent:
         return dm_np, delay_links, delay_nodes, delay_unit, list(env.flows), \
             float(loss_fn.detach()), loss_mse
 
-    # -- replay (reference :141-169) --------------------------------------------
     # -- scalar logging (reference log_init/log_scalar, :455-469) ---------------
     def log_init(self, logdir="logs"):
         import datetime
@@ -279,6 +278,7 @@ class ACOAgent:
         w = self.test_summary_writer if test else self.train_summary_writer
         w.scalar(name, variable, step)
 
+    # -- replay memory (reference :141-169) -------------------------------------
     def memorize(self, grads, loss, reward):
         self.memory.append((grads, loss, reward))
 

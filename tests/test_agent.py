@@ -268,3 +268,33 @@ def test_log_scalar_jsonl(tmp_path):
     row = json.loads(open([l for l in logs if l.name == "train.jsonl"][0]
                           ).readline())
     assert row == {"name": "loss", "value": 1.5, "step": 3}
+
+
+def test_chebconv_k3_matches_dense_polynomials(small_case):
+    """K=3 layer math vs dense Chebyshev polynomials of the extended
+    adjacency: sum_k T_k(A) X W_k + b with T2 = 2A·T1 - T0 (the GPU path
+    for K>=3 falls back to these torch layers)."""
+    from multihop_offload_amd.models.chebconv import ChebConvLayer
+    from multihop_offload_amd.queueing import ConflictCSR
+
+    g = small_case
+    ext = g.ext
+    Ee = ext.num_edges_ext
+    support = ConflictCSR(ext.ext_indptr, ext.ext_indices, device="cpu")
+    layer = ChebConvLayer(4, 6, K=3, dtype=torch.float64,
+                          gen=torch.Generator().manual_seed(0))
+    x = torch.randn(Ee, 4, dtype=torch.float64,
+                    generator=torch.Generator().manual_seed(1))
+    got = layer(x, support)
+
+    A = np.zeros((Ee, Ee))
+    for r in range(Ee):
+        lo, hi = ext.ext_indptr[r], ext.ext_indptr[r + 1]
+        A[r, ext.ext_indices[lo:hi]] = 1.0
+    assert np.allclose(A, A.T)
+    At = torch.as_tensor(A, dtype=torch.float64)
+    T0, T1 = x, At @ x
+    T2 = 2.0 * (At @ T1) - T0
+    want = (T0 @ layer.weight[0] + T1 @ layer.weight[1]
+            + T2 @ layer.weight[2] + layer.bias)
+    assert torch.allclose(got, want, atol=1e-10)

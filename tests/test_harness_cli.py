@@ -216,3 +216,28 @@ def test_train_batched_resume(tmp_path):
     ckpt_io.load(m_after, latest_checkpoint(d))
     for a, b in zip(m_after.parameters(), m_saved.parameters()):
         assert torch.allclose(a, b, atol=1e-12)
+
+
+def test_full_pipeline_datagen_train_test_figures(tmp_path):
+    """The reference's complete workflow end to end on fresh data:
+    data generation -> AdHoc_train -> AdHoc_test -> the three paper
+    figures, all through the compatible CLIs."""
+    from multihop_offload_amd import datagen
+    from multihop_offload_amd.harness import adhoc_train, adhoc_test, figures
+
+    datadir = str(tmp_path / "data")
+    datagen.main(["--datapath", datadir, "--size", "2", "--sizes", "20",
+                  "--seed", "900"])
+    mats = [f for f in os.listdir(datadir) if f.endswith(".mat")]
+    assert len(mats) == 2
+
+    out = str(tmp_path / "out")
+    common_args = ["--datapath", datadir, "--out", out,
+                   "--model_root", str(tmp_path / "model"),
+                   "--training_set", "PIPE", "--instances", "2",
+                   "--seed", "3", "--device", "cpu"]
+    adhoc_train.main(common_args + ["--epochs", "1", "--batch", "2"])
+    adhoc_test.main(common_args)
+    tcsv = glob.glob(os.path.join(out, "Adhoc_test_*.csv"))[0]
+    figures.main(["--csv", tcsv, "--fig_dir", str(tmp_path / "fig")])
+    assert len(list((tmp_path / "fig").glob("*.pdf"))) == 3
