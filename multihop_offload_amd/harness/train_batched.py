@@ -154,6 +154,9 @@ def main(argv=None):
                          "training tau the guard tracks")
     ap.add_argument("--eval_rounds", type=int, default=4)
     ap.add_argument("--eval_seed", type=int, default=12345)
+    ap.add_argument("--torch_profile", type=int, default=0,
+                    help="profile N steps with torch.profiler after warmup "
+                         "and write a chrome trace next to the model dir")
     ap.add_argument("--resume", action="store_true",
                     help="load the latest checkpoint from the model dir "
                          "before training (reference resume protocol)")
@@ -224,7 +227,19 @@ def main(argv=None):
     best_eval_params = None
     loads = ([float(x) for x in args.arrival_scales.split(",")]
              if args.arrival_scales else [args.arrival_scale])
+    profiler = None
+    profiler_on = False
+    if args.torch_profile and rank == 0:
+        from torch.profiler import profile, ProfilerActivity
+        acts = [ProfilerActivity.CPU]
+        if device.startswith("cuda"):
+            acts.append(ProfilerActivity.CUDA)
+        profiler = profile(activities=acts)
     for step in range(1, args.steps + 1):
+        if profiler is not None and not profiler_on and \
+                step >= max(args.steps - args.torch_profile, 1):
+            profiler.__enter__()
+            profiler_on = True
         engine = engines[step % len(engines)]
         jobs = engine.sample_jobs(loads[step % len(loads)], gen)
         for p in engine.model.parameters():
@@ -314,6 +329,12 @@ def main(argv=None):
             ckpt_io.save(engine.model,
                          os.path.join(actor_dir,
                                       f"cp-{step // args.save_every:04d}.ckpt"))
+    if profiler_on:
+        profiler.__exit__(None, None, None)
+        trace = os.path.join(args.model_root,
+                             f"torch_trace_{args.training_set}.json")
+        profiler.export_chrome_trace(trace)
+        print(f"torch.profiler trace -> {trace}", flush=True)
     if rank == 0:
         # ship the best-seen parameters: held-out-eval selection when
         # --eval_every was on, else the guard's best-training-tau snapshot

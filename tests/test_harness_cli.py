@@ -241,3 +241,19 @@ def test_full_pipeline_datagen_train_test_figures(tmp_path):
     tcsv = glob.glob(os.path.join(out, "Adhoc_test_*.csv"))[0]
     figures.main(["--csv", tcsv, "--fig_dir", str(tmp_path / "fig")])
     assert len(list((tmp_path / "fig").glob("*.pdf"))) == 3
+
+
+def test_train_batched_torch_profile(tmp_path):
+    """--torch_profile writes a chrome trace (SURVEY 5.1 tracing)."""
+    import json
+    from multihop_offload_amd.harness import train_batched
+    train_batched.main([
+        "--steps", "3", "--batch", "8", "--sizes", "20", "--distinct", "4",
+        "--workers", "0", "--seed", "5", "--device", "cpu",
+        "--guard_every", "0", "--save_every", "100", "--log_every", "100",
+        "--model_root", str(tmp_path), "--training_set", "PROF",
+        "--torch_profile", "2"])
+    trace = tmp_path / "torch_trace_PROF.json"
+    assert trace.exists()
+    blob = json.load(open(trace))
+    assert blob.get("traceEvents"), "empty trace"
