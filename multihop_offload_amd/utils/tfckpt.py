@@ -15,7 +15,6 @@ Only the features the reference checkpoints use are implemented
 
 from __future__ import annotations
 
-import os
 import struct
 from typing import Dict, Tuple
 
