@@ -71,6 +71,10 @@ class AgentConfig:
 
 
 class ACOAgent:
+    """Single-case actor-critic offloading agent with the reference's API
+    and training dynamics (see module docstring); the batched device path
+    is ``engine.EpisodeEngine``."""
+
     def __init__(self, cfg: AgentConfig, memory_size: int = 5000):
         self.cfg = cfg
         self.dtype = getattr(torch, cfg.dtype)

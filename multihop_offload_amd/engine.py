@@ -40,6 +40,8 @@ from .ops import dispatch as ops
 
 @dataclasses.dataclass
 class JobBatch:
+    """Padded per-graph job instances on device (the batched analog of
+    ``JobInstance``); padded slots are masked out of every reduction."""
     sources: torch.Tensor   # (B, J) int64, padded with a valid mobile node
     mask: torch.Tensor      # (B, J) bool — real jobs
     rates: torch.Tensor     # (B, J)
@@ -53,6 +55,8 @@ class JobBatch:
 
 @dataclasses.dataclass
 class EpisodeResult:
+    """Per-graph episode outcome (training losses populated when
+    ``train=True``)."""
     tau: torch.Tensor            # (B,) mean per-job empirical delay
     congest: torch.Tensor        # (B,) number of congested jobs
     num_jobs: torch.Tensor       # (B,)
@@ -62,6 +66,18 @@ class EpisodeResult:
 
 
 class EpisodeEngine:
+    """Device-resident batched episode engine — the MI355X product path.
+
+    Holds a batch of (same-N, ragged-E, per-graph-T) cases as flat
+    block-diagonal CSRs plus the index tables every step needs, and runs
+    full GNN/baseline/local episodes in a handful of fused HIP kernel
+    launches (torch fp64 composition on CPU, proven equal to the numpy
+    oracle in tests/test_engine.py).  Reference analogs:
+    ``offloading_v3.AdhocCloud`` decisions/evaluation +
+    ``gnn_offloading_agent.ACOAgent`` forward/backward, batched.
+    Mixed sizes: one engine per size, or ``CaseGraph.pad_to``.
+    """
+
     def __init__(self, cases: Sequence[CaseGraph], model: ChebConvStack,
                  device: str = "cpu", dtype: torch.dtype = torch.float32,
                  fp_iters: int = 10, walk_cap: Optional[int] = None):
