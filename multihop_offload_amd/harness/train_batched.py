@@ -206,9 +206,19 @@ def main(argv=None):
                                          dtype=dtype))
     engine = engines[0]
     dp.broadcast_params(engine.model)
-    opt = torch.optim.Adam(engine.model.parameters(),
-                           lr=args.learning_rate, eps=1e-7)
-    reducer = dp.FlatAllreduce(engine.model.parameters())
+    from ..ops import dispatch as mho_dispatch
+    use_fused = (device.startswith("cuda") and mho_dispatch.hip_available()
+                 and dtype == torch.float32)
+    if use_fused:
+        # one kernel: grad scale -> clipnorm -> Adam -> max_norm; flat_g is
+        # the single RCCL payload (same path bench.py measures)
+        from ..ops.functions import FusedAdam
+        opt = FusedAdam(engine.model, lr=args.learning_rate)
+        reducer = None
+    else:
+        opt = torch.optim.Adam(engine.model.parameters(),
+                               lr=args.learning_rate, eps=1e-7)
+        reducer = dp.FlatAllreduce(engine.model.parameters())
     gen = torch.Generator(device=device)
     gen.manual_seed(args.seed * 7919 + rank)
 
@@ -242,24 +252,36 @@ def main(argv=None):
             profiler_on = True
         engine = engines[step % len(engines)]
         jobs = engine.sample_jobs(loads[step % len(loads)], gen)
-        for p in engine.model.parameters():
-            p.grad = None
+        if use_fused:
+            opt.zero_grad()
+        else:
+            for p in engine.model.parameters():
+                p.grad = None
         res = engine.gnn_episode(jobs, explore=explore, gen=gen, train=True)
-        with torch.no_grad():
-            for p in engine.model.parameters():
-                if p.grad is not None:
-                    p.grad /= engine.B
-        reducer(average=True)
-        with torch.no_grad():
-            pre_norms = [float(p.grad.norm())
-                         for p in engine.model.parameters()
-                         if p.grad is not None]
-            for p in engine.model.parameters():
-                if p.grad is not None:
-                    n = p.grad.norm().clamp(min=1e-12)
-                    p.grad *= torch.clamp(n, max=1.0) / n
-        opt.step()
-        engine.model.apply_constraints()
+        if use_fused:
+            scale = 1.0 / (engine.B * world)
+            if world > 1:
+                torch.distributed.all_reduce(opt.flat_g)
+            with torch.no_grad():
+                pre_norms = [float(p.grad.norm()) * scale
+                             for p in engine.model.parameters()]
+            opt.step(scale=scale)
+        else:
+            with torch.no_grad():
+                for p in engine.model.parameters():
+                    if p.grad is not None:
+                        p.grad /= engine.B
+            reducer(average=True)
+            with torch.no_grad():
+                pre_norms = [float(p.grad.norm())
+                             for p in engine.model.parameters()
+                             if p.grad is not None]
+                for p in engine.model.parameters():
+                    if p.grad is not None:
+                        n = p.grad.norm().clamp(min=1e-12)
+                        p.grad *= torch.clamp(n, max=1.0) / n
+            opt.step()
+            engine.model.apply_constraints()
         explore = max(explore * args.explore_decay, 0.001)
         if args.lr_decay_at and step == args.lr_decay_at:
             for group in opt.param_groups:

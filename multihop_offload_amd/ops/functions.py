@@ -159,3 +159,37 @@ class FusedAdam:
         ext.fused_adam(self.flat_p, self.flat_g, self.m, self.v, self.seg,
                        self.step_dev, scale, self.lr, self.betas[0],
                        self.betas[1], self.eps, self.constraints)
+
+    # -- snapshot/restore (guard rollbacks, resume) ---------------------------
+    def state_dict(self):
+        """Optimizer state snapshot (tensors cloned; device-agnostic
+        restore).  The parameter buffer is NOT included — parameters are
+        snapshotted by the caller like with torch optimizers."""
+        return {"m": self.m.detach().clone(),
+                "v": self.v.detach().clone(),
+                "step_dev": self.step_dev.detach().clone(),
+                "lr": self.lr}
+
+    def load_state_dict(self, state):
+        with torch.no_grad():
+            self.m.copy_(state["m"])
+            self.v.copy_(state["v"])
+            self.step_dev.copy_(state["step_dev"])
+        self.lr = state["lr"]
+
+    @property
+    def param_groups(self):
+        """torch-optimizer-shaped lr access (the trainer's lr schedules
+        mutate ``group["lr"]``)."""
+        return [_FusedAdamGroup(self)]
+
+
+class _FusedAdamGroup(dict):
+    def __init__(self, opt):
+        super().__init__(lr=opt.lr)
+        self._opt = opt
+
+    def __setitem__(self, key, value):
+        super().__setitem__(key, value)
+        if key == "lr":
+            self._opt.lr = value
