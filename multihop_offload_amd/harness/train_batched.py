@@ -328,7 +328,13 @@ def main(argv=None):
             tau = float(torch.nanmean(res.tau))
             congest = int(res.congest.sum())
             njobs = int(res.num_jobs.sum())
-            gnorm = float(sum((p.grad ** 2).sum()
+            if use_fused:
+                # clipping happens inside the kernel: reconstruct the
+                # post-clip global norm from the scaled pre-clip norms
+                gnorm = float(sum(min(n, 1.0) ** 2
+                                  for n in pre_norms) ** 0.5)
+            else:
+                gnorm = float(sum((p.grad ** 2).sum()
                               for p in engine.model.parameters()
                               if p.grad is not None) ** 0.5)
             # pre-clip norms are the stability signal: the post-clip norm
