@@ -437,3 +437,47 @@ def test_engine_runner_gpu_matches_cpu_oracle():
     assert runner.engine.use_hip
     d_engine = runner.run_method("GNN", jobs, 0.0)
     np.testing.assert_allclose(d_engine, d_oracle, rtol=1e-3, atol=1e-2)
+
+
+@needs_gpu
+def test_fused_adam_state_dict_roundtrip():
+    """FusedAdam state snapshot/restore (the guard's rollback unit):
+    restoring params+state after divergent extra steps reproduces the
+    checkpoint trajectory exactly."""
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.ops.functions import FusedAdam
+
+    def make():
+        m = ChebConvStack(K=2, dtype=torch.float32, seed=8).to("cuda")
+        return m, FusedAdam(m, lr=1e-3)
+
+    torch.manual_seed(0)
+    m1, o1 = make()
+    grads = [torch.randn_like(o1.flat_g) for _ in range(4)]
+
+    def apply(o, g):
+        o.zero_grad()
+        o.flat_g.add_(g)
+        o.step(scale=0.5)
+
+    apply(o1, grads[0])
+    snap_p = o1.flat_p.detach().clone()
+    snap_s = o1.state_dict()
+    apply(o1, grads[1])          # divergent step
+    apply(o1, grads[2])
+    # rollback
+    with torch.no_grad():
+        o1.flat_p.copy_(snap_p)
+    o1.load_state_dict(snap_s)
+    o1.param_groups[0]["lr"] = o1.param_groups[0]["lr"]   # shim no-op
+    apply(o1, grads[3])
+    after_rollback = o1.flat_p.detach().clone()
+
+    # fresh trajectory: step0 then step3 directly
+    torch.manual_seed(0)
+    m2, o2 = make()
+    apply(o2, grads[0])
+    apply(o2, grads[3])
+    torch.cuda.synchronize()
+    torch.testing.assert_close(after_rollback, o2.flat_p,
+                               rtol=1e-6, atol=1e-7)
