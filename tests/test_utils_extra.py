@@ -111,3 +111,33 @@ def test_force_torch_escape_hatch(monkeypatch):
         w[0, i, (i + 1) % 4] = w[0, (i + 1) % 4, i] = 1.0
     d = dispatch.floyd_warshall(w)
     assert d[0, 0, 2] == 2.0
+
+
+def test_fused_adam_state_and_lr_shim_cpu():
+    """FusedAdam bookkeeping that needs no kernel: flat views, zero_grad,
+    state_dict roundtrip, and the param_groups lr shim the trainer's lr
+    schedules mutate."""
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.ops.functions import FusedAdam
+
+    m = ChebConvStack(K=2, dtype=torch.float32, seed=1)
+    opt = FusedAdam(m, lr=1e-4)
+    # params/grads are views of the flat buffers
+    p0 = next(m.parameters())
+    assert p0.data.data_ptr() >= opt.flat_p.data_ptr()
+    p0.grad.fill_(3.0)
+    assert float(opt.flat_g.abs().sum()) > 0
+    opt.zero_grad()
+    assert float(opt.flat_g.abs().sum()) == 0.0
+    # lr shim
+    g = opt.param_groups[0]
+    assert g["lr"] == 1e-4
+    g["lr"] /= 4.0
+    assert opt.lr == 2.5e-5
+    # state roundtrip
+    opt.m.fill_(0.5); opt.v.fill_(0.25); opt.step_dev.fill_(7)
+    s = opt.state_dict()
+    opt.m.zero_(); opt.v.zero_(); opt.step_dev.zero_(); opt.lr = 1.0
+    opt.load_state_dict(s)
+    assert float(opt.m.mean()) == 0.5 and float(opt.v.mean()) == 0.25
+    assert int(opt.step_dev) == 7 and opt.lr == 2.5e-5
