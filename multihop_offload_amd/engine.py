@@ -469,6 +469,11 @@ class EpisodeEngine:
         ``sp``: (B,N,N) shortest-path delays with ZERO diagonal;
         ``uds``: (B,N) per-node unit processing delays (inf at relays)."""
         B, J, S = self.B, self.Jmax, self.S
+        if S == 0:
+            # no servers anywhere in the batch: every job computes locally
+            # (the reference's cost vector degenerates to [local])
+            local = uds.gather(1, jobs.sources) * jobs.ul
+            return jobs.sources.clone(), local
         if self.use_hip and not prob:
             from .ops import dispatch
             ext = dispatch.require_hip()

@@ -439,3 +439,25 @@ def test_prob_mode_softmax_statistics():
         idx = (list(servers).index(d0) if d0 in servers else len(servers))
         counts[idx] += 1
     np.testing.assert_allclose(counts / trials, p_want, atol=0.03)
+
+
+def test_zero_server_batch_all_local():
+    """A batch whose cases have no servers degenerates to local computing
+    everywhere (and must not crash the decision/evaluation/critic path)."""
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+
+    rng = np.random.RandomState(0)
+    g = CaseGraph(10, seed=4, gtype="ba")
+    g.links_init(50.0, rng=rng)
+    for v in range(10):
+        g.set_mobile_bw(v, 10.0)
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, rng)
+    m = ChebConvStack(K=2, dtype=torch.float64, seed=0)
+    _wake(m)
+    eng = EpisodeEngine([g], m, device="cpu", dtype=torch.float64)
+    jb = eng.pack_jobs([jobs])
+    res = eng.gnn_episode(jb, train=True)
+    rl = eng.local_episode(jb)
+    np.testing.assert_allclose(res.delay_emp[0, :jobs.num_jobs],
+                               rl.delay_emp[0, :jobs.num_jobs], rtol=1e-12)
+    assert res.loss_fn is not None and torch.isfinite(res.loss_fn)
