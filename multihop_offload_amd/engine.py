@@ -163,6 +163,8 @@ class EpisodeEngine:
 
         # mobiles
         mob_counts = [len(c.mobile_nodes) for c in cases]
+        assert min(mob_counts) >= 1, \
+            "every case needs at least one mobile node (job sources)"
         self.Jmax = max(mob_counts) - 1 if max(mob_counts) > 1 else 1
         self.mobile_mask = self.roles == 0
 

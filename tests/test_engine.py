@@ -461,3 +461,18 @@ def test_zero_server_batch_all_local():
     np.testing.assert_allclose(res.delay_emp[0, :jobs.num_jobs],
                                rl.delay_emp[0, :jobs.num_jobs], rtol=1e-12)
     assert res.loss_fn is not None and torch.isfinite(res.loss_fn)
+
+
+def test_engine_rejects_caseless_mobiles():
+    """A case with no mobile nodes cannot source jobs — loud error at
+    engine construction, not a silent mis-sample later."""
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    g = CaseGraph(6, seed=1, gtype="ba")
+    g.links_init(50.0, rng=np.random.RandomState(0))
+    for v in range(3):
+        g.add_server(v, 300.0)
+    for v in range(3, 6):
+        g.add_relay(v)
+    m = ChebConvStack(K=2, dtype=torch.float64, seed=0)
+    with pytest.raises(AssertionError, match="mobile"):
+        EpisodeEngine([g], m, device="cpu", dtype=torch.float64)
