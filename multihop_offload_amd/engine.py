@@ -814,9 +814,27 @@ class EpisodeEngine:
             grad_dist, loss_mse = self.grad_dist_matrix(
                 grad_edge, dm, unit_mtx, written)
             self.per_sample_request = per_sample
-            dm.backward(grad_dist)
-            if per_sample:
-                self.last_per_sample_grads = self._collect_per_sample_grads()
+            if per_sample and not self.use_hip:
+                # CPU/torch fallback: instances are independent blocks, so
+                # a masked cotangent per graph yields its gradient set
+                params = list(self.model.parameters())
+                out = []
+                for b in range(self.B):
+                    gd = torch.zeros_like(grad_dist)
+                    gd[b] = grad_dist[b]
+                    gs = torch.autograd.grad(dm, params, grad_outputs=gd,
+                                             retain_graph=True,
+                                             allow_unused=True)
+                    out.append([torch.zeros_like(p) if g is None
+                                else g.detach()
+                                for p, g in zip(params, gs)])
+                self.last_per_sample_grads = out
+                dm.backward(grad_dist)
+            else:
+                dm.backward(grad_dist)
+                if per_sample:
+                    self.last_per_sample_grads = \
+                        self._collect_per_sample_grads()
             self.per_sample_request = False
 
         nj = jobs.mask.sum(1)

@@ -14,7 +14,9 @@ batched engine for the heavy lifting:
     reference's ``replay``;
   * checkpoints land in the compatible ``model_ChebConv_*`` layout.
 
-GPU-only (the per-sample path needs the fused ChebConv kernels).
+On GPU the per-sample sets come from the fused ChebConv backward's
+per-graph partials; on CPU from masked-cotangent backwards (slower,
+used for testing and small runs).
 """
 
 from __future__ import annotations
@@ -62,10 +64,12 @@ def main(argv=None):
     ap.add_argument("--save_every", type=int, default=500)
     ap.add_argument("--log_every", type=int, default=50)
     ap.add_argument("--init_scale", type=float, default=0.01)
+    ap.add_argument("--device", type=str, default=None)
     args = ap.parse_args(argv)
 
-    assert torch.cuda.is_available(), "train_replay needs a GPU"
-    model = ChebConvStack(K=args.K, dtype=torch.float32, seed=args.seed)
+    device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
+    dtype = torch.float32 if device.startswith("cuda") else torch.float64
+    model = ChebConvStack(K=args.K, dtype=dtype, seed=args.seed)
     with torch.no_grad():
         for layer in model.layers:
             layer.weight.mul_(args.init_scale)
@@ -76,10 +80,10 @@ def main(argv=None):
         build_training_cases(n, max(args.batch // len(sizes), 8),
                              args.distinct, args.T,
                              args.seed + 17 * n, workers=args.workers),
-        model, device="cuda", dtype=torch.float32) for n in sizes]
+        model, device=device, dtype=dtype) for n in sizes]
     opt = torch.optim.Adam(model.parameters(), lr=args.learning_rate,
                            eps=1e-7)
-    gen = torch.Generator(device="cuda")
+    gen = torch.Generator(device=device)
     gen.manual_seed(args.seed * 977)
     rng = random.Random(args.seed)
 

@@ -476,3 +476,37 @@ def test_engine_rejects_caseless_mobiles():
     m = ChebConvStack(K=2, dtype=torch.float64, seed=0)
     with pytest.raises(AssertionError, match="mobile"):
         EpisodeEngine([g], m, device="cpu", dtype=torch.float64)
+
+
+def test_per_sample_grads_cpu_match_oracle():
+    """CPU per-sample gradient sets (masked-cotangent fallback) equal the
+    oracle agent's replay-memory units exactly (fp64)."""
+    from multihop_offload_amd.agent import ACOAgent, AgentConfig
+    from multihop_offload_amd.env import AdhocCloudEnv
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+
+    g1, g2 = _case(seed=41, n=20), _case(seed=43, n=20)
+    rng = np.random.RandomState(5)
+    j1 = JobInstance.sample(g1.mobile_nodes, 0.15, rng)
+    j2 = JobInstance.sample(g2.mobile_nodes, 0.15, rng)
+
+    model = ChebConvStack(K=2, dtype=torch.float64, seed=9)
+    _wake(model)
+    eng = EpisodeEngine([g1, g2], model, device="cpu", dtype=torch.float64)
+    res = eng.gnn_episode(eng.pack_jobs([j1, j2]), train=True,
+                          per_sample=True)
+    psg = eng.last_per_sample_grads
+    assert len(psg) == 2
+
+    agent = ACOAgent(AgentConfig(seed=9), 10)
+    _wake(agent.model)
+    for b, (g, j) in enumerate(((g1, j1), (g2, j2))):
+        env = AdhocCloudEnv(g)
+        env.set_jobs(j)
+        agent.forward_backward(env, 0.0, np.random.RandomState(0))
+        for got, want in zip(psg[b], agent.memory[-1][0]):
+            assert torch.allclose(got, want, atol=1e-10), \
+                (b, (got - want).abs().max())
+    # and the summed grads in p.grad equal the sum of the sets
+    for i, p in enumerate(model.parameters()):
+        assert torch.allclose(p.grad, psg[0][i] + psg[1][i], atol=1e-10)

@@ -257,3 +257,16 @@ def test_train_batched_torch_profile(tmp_path):
     assert trace.exists()
     blob = json.load(open(trace))
     assert blob.get("traceEvents"), "empty trace"
+
+
+def test_train_replay_cpu(tmp_path):
+    """The replay trainer (reference training dynamics) now runs on CPU
+    via the masked-cotangent per-sample fallback."""
+    from multihop_offload_amd.harness import train_replay
+    history = train_replay.main([
+        "--steps", "4", "--batch", "8", "--sizes", "20", "--distinct", "4",
+        "--workers", "0", "--seed", "6", "--device", "cpu",
+        "--replay_batch", "8", "--memory", "64", "--save_every", "100",
+        "--log_every", "2", "--model_root", str(tmp_path),
+        "--training_set", "RPLC"])
+    assert history, "no log records"
