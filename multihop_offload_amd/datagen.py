@@ -86,7 +86,8 @@ def generate_case(num_nodes: int, seed: int, gtype: str = "ba", m: int = 2,
                     nodes_info[part[i], 0] = 0
                     nodes_info[part[i], 1] = mb[i - k]
         else:
-            n_near = max(0, num_servers - len(partition[server_side]))
+            n_near = min(max(0, num_servers - len(partition[server_side])),
+                         len(part))
             if n_near > 0:
                 bws = (rng.pareto(2.0, n_near) + 1) * 100
                 for i in range(n_near):
@@ -97,6 +98,14 @@ def generate_case(num_nodes: int, seed: int, gtype: str = "ba", m: int = 2,
             for i in range(n_near, len(part)):
                 nodes_info[part[i], 0] = 0
                 nodes_info[part[i], 1] = mb[i - n_near]
+
+    # guarantee at least one job source: tiny graphs with a large relay
+    # cut can otherwise end up all-server/all-relay
+    if not np.any(nodes_info[:, 0] == 0):
+        servers = np.nonzero(nodes_info[:, 0] == 1)[0]
+        demote = servers[np.argmin(nodes_info[servers, 1])]
+        nodes_info[demote, 0] = 0
+        nodes_info[demote, 1] = (rng.pareto(2.0) + 1) * 8
 
     return {
         "network": {"num_nodes": num_nodes, "seed": seed, "m": m,

@@ -303,7 +303,7 @@ def test_walk_overflow_guard():
     assert int(nhop.max()) <= 1
 
 
-@pytest.mark.parametrize("seed", range(8))
+@pytest.mark.parametrize("seed", range(12))
 def test_engine_oracle_sweep_random_topologies(seed):
     """Randomized hardening sweep: for random topologies (mixed families),
     random role assignments from the datagen distributions, and random job
