@@ -137,3 +137,38 @@ def test_delay_fallback_tensor_T():
     d = delay_with_fallback(lam, mu, T, 101.0)
     assert np.isclose(d[0].item(), 700.0 * 10 / (101 * 5))
     assert np.isclose(d[1].item(), 1000.0 * 10 / (101 * 5))
+
+
+def test_fixed_point_properties_random():
+    """Analytic invariants of the contention fixed point over random
+    conflict graphs: 0 < mu <= rates/(1) bounds, mu <= mu0 start,
+    and adding load never increases any mu (monotonicity)."""
+    from multihop_offload_amd.queueing import ConflictCSR, fixed_point_mu
+
+    for seed in range(6):
+        rng = np.random.RandomState(200 + seed)
+        E = rng.randint(4, 40)
+        # random symmetric conflict lists
+        adj = np.triu(rng.rand(E, E) < 0.3, 1)
+        adj = adj | adj.T
+        indptr = np.zeros(E + 1, dtype=np.int64)
+        indices = []
+        for l in range(E):
+            nbs = np.nonzero(adj[l])[0]
+            indices.extend(nbs.tolist())
+            indptr[l + 1] = len(indices)
+        conf = ConflictCSR(indptr, np.asarray(indices, dtype=np.int64),
+                           device="cpu")
+        rates = torch.as_tensor(rng.uniform(5.0, 100.0, E))
+        cf_degs = torch.as_tensor(adj.sum(1).astype(np.float64))
+        lam = torch.as_tensor(rng.uniform(0.0, 10.0, E))
+
+        mu = fixed_point_mu(lam, rates, cf_degs, conf, 10)
+        assert torch.all(mu > 0)
+        assert torch.all(mu <= rates + 1e-12)
+        # more load somewhere -> no mu increases anywhere
+        lam2 = lam.clone()
+        j = rng.randint(E)
+        lam2[j] += 5.0
+        mu2 = fixed_point_mu(lam2, rates, cf_degs, conf, 10)
+        assert torch.all(mu2 <= mu + 1e-9), (mu2 - mu).max()
