@@ -253,7 +253,11 @@ class ACOAgent:
         du[np.isinf(du)] = np.nan
         with np.errstate(invalid="ignore"):
             diff = dm_np - du
-        loss_mse = float(np.nanmean(diff ** 2))
+        sq = diff ** 2
+        valid = ~np.isnan(sq)
+        # nanmean semantics without the all-NaN RuntimeWarning (zero-job
+        # instances have an empty unit matrix)
+        loss_mse = float(sq[valid].mean()) if valid.any() else float("nan")
         grad_dist += np.nan_to_num(0.001 * diff, nan=0.0)
 
         # ---- actor VJP (reference :448)
