@@ -244,7 +244,7 @@ def main(argv=None):
         acts = [ProfilerActivity.CPU]
         if device.startswith("cuda"):
             acts.append(ProfilerActivity.CUDA)
-        profiler = profile(activities=acts)
+        profiler = profile(activities=acts, acc_events=True)
     for step in range(1, args.steps + 1):
         if profiler is not None and not profiler_on and \
                 step >= max(args.steps - args.torch_profile, 1):
