@@ -298,3 +298,15 @@ def test_chebconv_k3_matches_dense_polynomials(small_case):
     want = (T0 @ layer.weight[0] + T1 @ layer.weight[1]
             + T2 @ layer.weight[2] + layer.bias)
     assert torch.allclose(got, want, atol=1e-10)
+
+
+def test_learning_decay_schedule(small_case, jobs_for):
+    """learning_decay follows Keras ExponentialDecay(decay_steps=100,
+    staircase=False): lr = lr0 * rate^(step/100), applied per Adam apply."""
+    from multihop_offload_amd.agent import AgentConfig
+    agent = ACOAgent(AgentConfig(seed=0, learning_decay=0.9), 50)
+    zero = [torch.zeros_like(p) for p in agent.model.parameters()]
+    for k in range(5):
+        agent._apply_one(zero)
+        want = 1e-4 * 0.9 ** ((k) / 100.0)   # lr set BEFORE the k-th apply
+        assert np.isclose(agent.optimizer.param_groups[0]["lr"], want)
