@@ -153,6 +153,10 @@ def main(argv=None):
                          "parameters, a stronger selection signal than the "
                          "training tau the guard tracks")
     ap.add_argument("--eval_rounds", type=int, default=4)
+    ap.add_argument("--eval_on_train", action="store_true",
+                    help="evaluate on the training topologies instead of "
+                         "the default held-out ones (held-out catches "
+                         "topology overfitting — see docs/TRAINING.md)")
     ap.add_argument("--eval_seed", type=int, default=12345)
     ap.add_argument("--torch_profile", type=int, default=0,
                     help="profile N steps with torch.profiler after warmup "
@@ -222,6 +226,17 @@ def main(argv=None):
     gen = torch.Generator(device=device)
     gen.manual_seed(args.seed * 7919 + rank)
 
+    eval_engines = engines
+    if args.eval_every and not args.eval_on_train:
+        # held-out topologies (unseen seeds): training-set eval hides
+        # topology overfitting (docs/TRAINING.md, LONGSOAK exhibit)
+        eval_engines = [EpisodeEngine(
+            build_training_cases(n, min(per_size, 16),
+                                 min(args.distinct, 16), args.T,
+                                 args.seed + 900000 + 17 * n,
+                                 workers=args.workers),
+            model, device=device, dtype=dtype) for n in sizes]
+
     actor_dir = model_dir(args.model_root, args.training_set)
     explore = args.explore
     t0 = time.time()
@@ -288,7 +303,7 @@ def main(argv=None):
                 group["lr"] *= 0.1
         if args.eval_every and step % args.eval_every == 0:
             eval_tau = _dist_mean(
-                evaluate_policy(engines, loads, args.eval_seed,
+                evaluate_policy(eval_engines, loads, args.eval_seed,
                                 args.eval_rounds),
                 world, engine.device)
             if eval_tau < best_eval_tau:

@@ -84,7 +84,9 @@ def test_train_batched_eval_selection(tmp_path):
     ckpt_io.load(m, str(tmp_path / "model_ChebConv_EVT_a5_c5_ACO_agent"
                         / "cp-9999.ckpt"))
     # reproduce the eval at the shipped parameters: matches best_eval_tau
-    cases = train_batched.build_training_cases(20, 8, 4, 1000, 5 + 17 * 20,
+    # (held-out eval topologies: seed + 900000 + 17*n)
+    cases = train_batched.build_training_cases(20, 8, 4, 1000,
+                                               5 + 900000 + 17 * 20,
                                                workers=0)
     from multihop_offload_amd.engine import EpisodeEngine
     eng = EpisodeEngine(cases, m, device="cpu", dtype=torch.float64)
