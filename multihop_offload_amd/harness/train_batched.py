@@ -43,6 +43,10 @@ def build_training_cases(n_nodes, batch, distinct, T, seed, gtype="ba",
     distributions, optionally generated in parallel), replicated to `batch`
     cases with independent link-rate draws (structure shared)."""
     from ..graphs import CaseGraph
+    if distinct > batch:
+        # bases beyond the batch size would never be instantiated — cap
+        # (a batch of B cases can carry at most B distinct topologies)
+        distinct = batch
     rng = np.random.RandomState(seed)
     tasks = [(n_nodes, seed + d, gtype) for d in range(distinct)]
     if workers > 1:
