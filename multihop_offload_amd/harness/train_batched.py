@@ -157,6 +157,10 @@ def main(argv=None):
                          "parameters, a stronger selection signal than the "
                          "training tau the guard tracks")
     ap.add_argument("--eval_rounds", type=int, default=4)
+    ap.add_argument("--eval_cases", type=int, default=0,
+                    help="held-out eval cases per size (0 = min(per-size "
+                         "batch, 16)); larger sets give a less noisy "
+                         "selection signal")
     ap.add_argument("--eval_on_train", action="store_true",
                     help="evaluate on the training topologies instead of "
                          "the default held-out ones (held-out catches "
@@ -234,9 +238,10 @@ def main(argv=None):
     if args.eval_every and not args.eval_on_train:
         # held-out topologies (unseen seeds): training-set eval hides
         # topology overfitting (docs/TRAINING.md, LONGSOAK exhibit)
+        ec = args.eval_cases or min(per_size, 16)
+        ed = args.eval_cases or min(args.distinct, 16)
         eval_engines = [EpisodeEngine(
-            build_training_cases(n, min(per_size, 16),
-                                 min(args.distinct, 16), args.T,
+            build_training_cases(n, ec, ed, args.T,
                                  args.seed + 900000 + 17 * n,
                                  workers=args.workers),
             model, device=device, dtype=dtype) for n in sizes]
