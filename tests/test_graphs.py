@@ -232,3 +232,28 @@ def test_cf_radius_engine_matches_oracle():
                                rtol=1e-9)
     for p, go in zip(model2.parameters(), agent.memory[-1][0]):
         assert torch.allclose(p.grad, go, atol=1e-9)
+
+
+def test_all_committed_mat_cases_load():
+    """Every committed demo .mat case loads through the reference schema
+    with consistent roles and a connected topology (C10 dataset check)."""
+    import glob
+    from scipy.sparse import csr_matrix
+    from scipy.sparse.csgraph import connected_components
+
+    mats = sorted(glob.glob("data_samples/**/*.mat", recursive=True))
+    if not mats:
+        import pytest
+        pytest.skip("no committed .mat samples")
+    assert len(mats) >= 10
+    for path in mats:
+        g = CaseGraph.from_mat(path)
+        assert g.num_links == len(g.mat_link_rate)
+        assert len(g.servers) >= 1 and len(g.mobile_nodes) >= 1
+        ncomp, _ = connected_components(
+            csr_matrix(g.adj.astype(float)), directed=False)
+        assert ncomp == 1, path
+        g.links_init(g.mat_link_rate, rng=np.random.RandomState(0))
+        assert (g.link_rates >= 0).all()
+        ext = g.ext
+        assert ext.num_edges_ext == g.num_links + len(ext.comp_nodes)
