@@ -510,3 +510,43 @@ def test_per_sample_grads_cpu_match_oracle():
     # and the summed grads in p.grad equal the sum of the sets
     for i, p in enumerate(model.parameters()):
         assert torch.allclose(p.grad, psg[0][i] + psg[1][i], atol=1e-10)
+
+
+def test_engine_after_topology_update():
+    """Mobility: after random_walk + topology_update the rebuilt engine
+    still matches the oracle on the NEW topology (CSR tables, link ids
+    and conflict graph all refreshed consistently)."""
+    from multihop_offload_amd.agent import ACOAgent, AgentConfig
+    from multihop_offload_amd.env import AdhocCloudEnv
+    from multihop_offload_amd.harness.common import run_method
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+
+    rng = np.random.RandomState(3)
+    g = CaseGraph(25, seed=2, gtype="poisson", m=6)
+    g.links_init(50.0, rng=rng)
+    g.add_relay(0)
+    for s in (2, 3, 4):
+        g.add_server(s, 300.0)
+    for v in range(25):
+        if g.roles[v] == 0:
+            g.set_mobile_bw(v, 10.0)
+
+    adj, pos = g.random_walk(ss=0.08, n=6, rng=rng)
+    g.topology_update(adj, pos)
+    g.links_init(50.0, rng=rng)          # fresh rates on the new links
+    jobs = JobInstance.sample(g.mobile_nodes, 0.15, rng)
+
+    agent = ACOAgent(AgentConfig(seed=6), 10)
+    _wake(agent.model)
+    env = AdhocCloudEnv(g)
+    env.set_jobs(jobs)
+    d_oracle = run_method("GNN", agent, env, 0.0, np.random.RandomState(0))
+
+    model = ChebConvStack(K=2, dtype=torch.float64, seed=6)
+    _wake(model)
+    eng = EpisodeEngine([g], model, device="cpu", dtype=torch.float64)
+    res = eng.gnn_episode(eng.pack_jobs([jobs]), train=True)
+    np.testing.assert_allclose(res.delay_emp[0, :jobs.num_jobs], d_oracle,
+                               rtol=1e-9)
+    for p, go in zip(model.parameters(), agent.memory[-1][0]):
+        assert torch.allclose(p.grad, go, atol=1e-9)
