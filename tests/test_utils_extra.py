@@ -141,3 +141,15 @@ def test_fused_adam_state_and_lr_shim_cpu():
     opt.load_state_dict(s)
     assert float(opt.m.mean()) == 0.5 and float(opt.v.mean()) == 0.25
     assert int(opt.step_dev) == 7 and opt.lr == 2.5e-5
+
+
+def test_figures_from_committed_results(tmp_path):
+    """The committed result CSVs (C11) still drive the three paper
+    figures."""
+    import glob
+    from multihop_offload_amd.harness import figures
+    csvs = glob.glob("out_samples/Adhoc_test_*.csv")
+    if not csvs:
+        pytest.skip("no committed test CSV")
+    figures.main(["--csv", csvs[0], "--fig_dir", str(tmp_path)])
+    assert len(list(tmp_path.glob("*.pdf"))) == 3
