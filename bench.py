@@ -46,10 +46,19 @@ def build_cases(n_nodes, batch, distinct, T, seed, gtype="ba", workers=None):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--steps", type=int, default=None,
+                    help="timed steps (default: 200 on GPU — long enough "
+                         "for SMI utilisation sampling to corroborate the "
+                         "timed region — 10 on CPU)")
+    ap.add_argument("--warmup", type=int, default=None)
     ap.add_argument("--batch", type=int, default=256,
-                    help="episodes per GPU per step")
+                    help="episodes per GPU per step (weak scaling)")
+    ap.add_argument("--global_batch", type=int, default=0,
+                    help="TOTAL episodes per step split across ranks "
+                         "(strong scaling; overrides --batch).  BASELINE "
+                         "config 3 is `--gpus 8 --global_batch 256 "
+                         "--nodes 110`: a batch of 256 110-node graphs "
+                         "sharded over 8 GPUs")
     ap.add_argument("--nodes", type=int, default=100)
     ap.add_argument("--distinct", type=int, default=16)
     ap.add_argument("--gtype", type=str, default="ba",
@@ -75,8 +84,16 @@ def main():
     from multihop_offload_amd.parallel import dp
 
     rank, world = dp.init_from_env()
+    if args.global_batch:
+        assert args.global_batch % world == 0, \
+            "--global_batch must divide evenly across ranks"
+        args.batch = args.global_batch // world
     use_cuda = torch.cuda.is_available()
     device = args.device or ("cuda" if use_cuda else "cpu")
+    if args.steps is None:
+        args.steps = 200 if device.startswith("cuda") else 10
+    if args.warmup is None:
+        args.warmup = 20 if device.startswith("cuda") else 3
     dtype = torch.float32 if device.startswith("cuda") else torch.float64
     import torch.distributed as dist
     distributed = dist.is_available() and dist.is_initialized()
@@ -198,7 +215,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong" if args.global_batch else "weak",
             "vs_baseline": value / 4.0,
             "dtype": "fp32" if dtype == torch.float32 else "fp64",
             "data": "synthetic",

@@ -10,6 +10,9 @@ broken as-is: stale ``from offloading import *`` and removed
     server side (smaller partition; Pareto(2)+1 ×100 bandwidths, sorted
     descending) and a mobile side (Pareto(2)+1 ×8);
   * link rates ~ U(30, 70);
+  * one documented deviation: leftover server-side nodes get mobile
+    Pareto(2)×8 bandwidths instead of the reference's role-0/bandwidth-0
+    (see the inline comment in :func:`generate_case`);
   * saved as ``aco_case_seed{S}_m{M}_n{N}_s{num_servers}.mat`` with keys
     {network{num_nodes,seed,m,gtype}, adj (sparse float), link_rate,
     nodes_info (N×2), pos_c (N×2)} (``data_generation_offloading.py:136-144``).
@@ -77,8 +80,15 @@ def generate_case(num_nodes: int, seed: int, gtype: str = "ba", m: int = 2,
             for i in range(k):
                 nodes_info[part[i], 0] = 1
                 nodes_info[part[i], 1] = bws[i]
-            # leftover nodes on the server side (when num_servers < side
-            # size) stay mobile with Pareto×8 bandwidths
+            # DELIBERATE DEVIATION from the reference
+            # (data_generation_offloading.py:90-133): leftover nodes on the
+            # server side (when num_servers < side size) become mobiles
+            # with Pareto(2)×8 bandwidths; the reference leaves them as
+            # role-0 nodes with bandwidth 0, which makes them job sources
+            # that can never compute anything.  Cross-comparisons against
+            # reference-GENERATED datasets should account for this shifted
+            # role/bandwidth distribution (reference-SHIPPED `.mat` files
+            # load unchanged through CaseGraph.from_mat).
             n_rest = len(part) - k
             if n_rest > 0:
                 mb = (rng.pareto(2.0, n_rest) + 1) * 8

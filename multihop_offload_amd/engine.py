@@ -241,6 +241,12 @@ class EpisodeEngine:
         self._bidx = torch.arange(B, device=self.device)
         self._eye = torch.eye(N, dtype=torch.bool,
                               device=self.device)[None, :, :]
+        # truncated-walk counter, accumulated on device (sync-free inner
+        # loop); callers poll check_overflow() at log points — a greedy
+        # walk that fails to reach its destination within walk_cap hops
+        # would otherwise silently underestimate loads/delays
+        self.overflow_total = torch.zeros((), dtype=torch.int64,
+                                          device=self.device)
         # per-graph horizon T (BASELINE config 5: ragged batches may mix T)
         self.T_arr = t([float(c.T) for c in cases])            # (B,)
         self.T_link = self.T_arr.repeat_interleave(E)          # (B*E,)
@@ -577,6 +583,9 @@ class EpisodeEngine:
         else:
             route_links = torch.full((B, J, 0), -1, dtype=torch.int64,
                                      device=self.device)
+        # jobs still short of their destination after walk_cap hops were
+        # truncated — count them (device-side; see check_overflow)
+        self.overflow_total += ((node != dst) & jobs.mask).sum()
         return route_links, hops
 
     # --------------------------------------------------------- evaluation
@@ -597,10 +606,30 @@ class EpisodeEngine:
                 self.proc_bws.contiguous(), self.k_edges,
                 self.T_arr.contiguous(), H, self.fp_iters)
             self._last_overflow = overflow
+            self.overflow_total += overflow.sum().to(torch.int64)
             return rl, nhop, delay_emp, unit_mtx, written
         rl, nhop = self.route_walk(jobs, dst, sp)
         delay_emp, unit_mtx, written, *_ = self.evaluate(jobs, dst, rl, nhop)
         return rl, nhop, delay_emp, unit_mtx, written
+
+    def check_overflow(self, strict: bool = True) -> int:
+        """Truncated-walk detection: number of greedy walks that failed to
+        reach their destination since the last call (synchronises; call at
+        log points, not inside the hot loop).  ``strict`` raises on any —
+        the evaluation harnesses use that; trainers log the count instead
+        (a transiently pathological delay matrix under exploration should
+        not kill a long run)."""
+        n = int(self.overflow_total)
+        if n:
+            self.overflow_total.zero_()
+            if strict:
+                raise RuntimeError(
+                    f"{n} greedy routing walk(s) failed to reach their "
+                    f"destination within walk_cap={self.walk_cap} hops — "
+                    "loads/delays for those jobs are truncated "
+                    "(pathological delay matrix or walk_cap too small "
+                    "for this topology)")
+        return n
 
     def evaluate(self, jobs: JobBatch, dst: torch.Tensor,
                  route_links: torch.Tensor, nhop: torch.Tensor):

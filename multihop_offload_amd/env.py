@@ -218,15 +218,14 @@ class AdhocCloudEnv:
         return link_delay, server_delay, unit_mtx
 
 
-_CONF_ROWS_CACHE = {}
-
-
 def _conf_rows(g: CaseGraph) -> np.ndarray:
-    key = id(g)
-    got = _CONF_ROWS_CACHE.get(key)
+    # Cached ON the graph instance (a module-level id(g)-keyed dict is
+    # unsafe: CPython reuses ids after garbage collection, so a long
+    # harness run could silently pick up another topology's row indices).
+    got = getattr(g, "_conf_rows_cache", None)
     if got is None or len(got) != len(g.conf_indices):
         got = np.repeat(np.arange(g.num_links), np.diff(g.conf_indptr))
-        _CONF_ROWS_CACHE[key] = got
+        g._conf_rows_cache = got
     return got
 
 

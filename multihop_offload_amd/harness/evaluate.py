@@ -59,6 +59,7 @@ def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
                     d["jobs"] += int(res.num_jobs.sum())
                     d["ratio_sum"] += float(r[ok].sum())
                     d["ratio_n"] += int(ok.sum())
+        engine.check_overflow()       # strict: truncated walks void an eval
         per_size[n] = {
             m: {"tau": d["tau_sum"] / max(d["tau_n"], 1),
                 "congest_ratio": d["congest"] / max(d["jobs"], 1),

@@ -365,6 +365,8 @@ def main(argv=None):
             # is pinned at sqrt(#tensors) whenever every tensor saturates
             # the per-tensor clip (see ROUND2.md stability probe)
             rec = {"step": step, "tau": tau,
+                   "walk_overflow": sum(e.check_overflow(strict=False)
+                                        for e in engines),
                    "grad_norm_preclip_max": max(pre_norms) if pre_norms
                    else 0.0,
                    "grad_norm_preclip_sum": float(np.hypot.reduce(

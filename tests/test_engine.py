@@ -299,8 +299,15 @@ def test_walk_overflow_guard():
     # force a far destination for every job
     far = torch.full_like(jobs.sources, int(g.servers[0]))
     rl, nhop = engine.route_walk(jobs, far, sp)
-    # capped at walk_cap hops; unreached jobs simply stop
+    # capped at walk_cap hops; truncation is detected via the overflow
+    # counter (strict check raises, non-strict returns the count)
     assert int(nhop.max()) <= 1
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError, match="walk_cap"):
+        engine.check_overflow()
+    assert engine.check_overflow(strict=False) == 0   # counter was reset
+    rl, nhop = engine.route_walk(jobs, far, sp)
+    assert engine.check_overflow(strict=False) > 0
 
 
 @pytest.mark.parametrize("seed", range(12))
