@@ -80,7 +80,8 @@ class EpisodeEngine:
 
     def __init__(self, cases: Sequence[CaseGraph], model: ChebConvStack,
                  device: str = "cpu", dtype: torch.dtype = torch.float32,
-                 fp_iters: int = 10, walk_cap: Optional[int] = None):
+                 fp_iters: int = 10, walk_cap: Optional[int] = None,
+                 delay_clamp: float = 0.0):
         assert len(cases) > 0
         N = cases[0].num_nodes
         T = cases[0].T
@@ -97,6 +98,9 @@ class EpisodeEngine:
         self.device = torch.device(device)
         self.dtype = dtype
         self.fp_iters = fp_iters
+        # pole mitigation for training: clamp the 1/(mu-lam) delay branch
+        # at this value (0 = off, reference semantics) — see queueing.py
+        self.delay_clamp = float(delay_clamp)
         self.model = model.to(self.device)
         self.walk_cap = walk_cap or N
 
@@ -241,6 +245,14 @@ class EpisodeEngine:
         self._bidx = torch.arange(B, device=self.device)
         self._eye = torch.eye(N, dtype=torch.bool,
                               device=self.device)[None, :, :]
+        # device-side exploration scalar + stateless-RNG state for the
+        # in-kernel ε-greedy/softmax sampling (capture-safe: a hipGraph
+        # replay reads the CURRENT explore value and a counter this engine
+        # bumps with a device op each decision)
+        self._explore_buf = torch.zeros((), dtype=torch.float32,
+                                        device=self.device)
+        self._rng_state = torch.tensor([12345, 0], dtype=torch.int64,
+                                       device=self.device)
         # truncated-walk counter, accumulated on device (sync-free inner
         # loop); callers poll check_overflow() at log points — a greedy
         # walk that fails to reach its destination within walk_cap hops
@@ -328,6 +340,12 @@ class EpisodeEngine:
             assert dtype == torch.float32, "HIP kernels are fp32"
             from .ops import dispatch
             dispatch.require_hip()
+
+    def set_rng_seed(self, seed: int):
+        """Seed the in-kernel stateless sampler (per-DP-rank seeds keep the
+        ε-greedy draws independent across ranks)."""
+        self._rng_state[0] = int(seed)
+        self._rng_state[1] = 0
 
     # ------------------------------------------------------------------ jobs
     def sample_jobs(self, arrival_scale: float,
@@ -418,7 +436,16 @@ class EpisodeEngine:
                 for layer in self.model.layers:
                     params += [layer.weight, layer.bias]
                 lam = ChebStackFn.apply(x.contiguous(), self, *params)
-            else:  # unsupported model shape: torch layers on GPU
+            else:
+                # no silent eager fallback on GPU: unsupported model
+                # shapes refuse loudly unless explicitly overridden
+                import os as _os
+                if _os.environ.get("MHO_ALLOW_TORCH_GPU") != "1":
+                    raise RuntimeError(
+                        "model shape unsupported by the fused ChebConv "
+                        "kernels (feature width > 32 or dropout > 0); "
+                        "refusing eager torch on GPU — set "
+                        "MHO_ALLOW_TORCH_GPU=1 to override")
                 lam = self.model(x.reshape(B * Ee, 4),
                                  self.support).reshape(B, Ee)
             dm = ActorHeadFn.apply(lam, self)
@@ -430,10 +457,11 @@ class EpisodeEngine:
         mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),
                             self.cf_degs.reshape(-1), self.conf,
                             self.fp_iters)
-        link_delay = delay_with_fallback(lam_link, mu, self.T_link, 101.0) \
-            .reshape(B, E)
+        link_delay = delay_with_fallback(lam_link, mu, self.T_link, 101.0,
+                                         self.delay_clamp).reshape(B, E)
         node_delay = delay_with_fallback(lam[:, E:], self.bw_comp,
-                                         self.T_arr[:, None], 100.0)
+                                         self.T_arr[:, None], 100.0,
+                                         self.delay_clamp)
 
         dm = self._delay_matrix(link_delay, node_delay)
         return dm, link_delay, node_delay
@@ -482,22 +510,29 @@ class EpisodeEngine:
             # (the reference's cost vector degenerates to [local])
             local = uds.gather(1, jobs.sources) * jobs.ul
             return jobs.sources.clone(), local
-        if self.use_hip and not prob:
+        if self.use_hip:
             from .ops import dispatch
             ext = dispatch.require_hip()
+            # ε-greedy + softmax sampling run INSIDE the kernel: explore is
+            # read from a device scalar (pass a 0-dim tensor for hipGraph
+            # capture — the replay sees the current value) and randomness
+            # comes from the stateless counter RNG, bumped here with a
+            # device op (capture-safe, advances per replay)
+            expl_t = None
+            if torch.is_tensor(explore):
+                expl_t = explore.to(self.device, torch.float32)
+            elif explore > 0:
+                self._explore_buf.fill_(float(explore))
+                expl_t = self._explore_buf
+            rng_t = None
+            if expl_t is not None or prob:
+                self._rng_state[1] += 1
+                rng_t = self._rng_state
             dst, islocal = ext.decide(
                 sp.contiguous(), self.sp_hop.contiguous(), uds.contiguous(),
                 self.k_servers, jobs.sources, jobs.mask,
-                jobs.ul.contiguous(), jobs.dl.contiguous())
-            if explore > 0:
-                nS = self.server_mask.sum(1, keepdim=True)
-                r = torch.rand(B, J, device=self.device, generator=gen)
-                rc = (torch.rand(B, J, device=self.device, generator=gen)
-                      * (nS + 1).to(self.dtype)).to(torch.int64)
-                dst_rand = torch.where(
-                    rc >= nS, jobs.sources,
-                    self.servers_safe.gather(1, rc.clamp(max=S - 1)))
-                dst = torch.where(jobs.mask & (r < explore), dst_rand, dst)
+                jobs.ul.contiguous(), jobs.dl.contiguous(),
+                expl_t, rng_t, 1 if prob else 0)
             return dst, None
 
         bJ = self._bidx[:, None]
@@ -732,7 +767,7 @@ class EpisodeEngine:
                 jobs.dl.contiguous(), self.k_conf_indptr, self.k_conf_base,
                 self.k_conf_cols, self.link_rates.contiguous(),
                 self.bw_comp.contiguous(), self.T_arr.contiguous(), Ee,
-                self.fp_iters)
+                self.fp_iters, self.delay_clamp)
             return grad_edge, loss.sum()
         H = route_links.shape[2]
         valid = route_links >= 0
@@ -761,10 +796,11 @@ class EpisodeEngine:
         mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),
                             self.cf_degs.reshape(-1), self.conf,
                             self.fp_iters)
-        link_d = delay_with_fallback(lam_link, mu, self.T_link, 101.0) \
-            .reshape(B, E)
+        link_d = delay_with_fallback(lam_link, mu, self.T_link, 101.0,
+                                     self.delay_clamp).reshape(B, E)
         node_d = delay_with_fallback(link_load[:, E:], self.bw_comp,
-                                     self.T_arr[:, None], 100.0)
+                                     self.T_arr[:, None], 100.0,
+                                     self.delay_clamp)
         unit_edge = torch.cat([link_d, node_d], dim=1)         # (B,Ē)
         delay_job_edge = torch.maximum(
             jobs_data * unit_edge[..., None] * routes, routes)

@@ -172,6 +172,18 @@ def main(argv=None):
     ap.add_argument("--resume", action="store_true",
                     help="load the latest checkpoint from the model dir "
                          "before training (reference resume protocol)")
+    ap.add_argument("--delay_clamp", type=float, default=0.0,
+                    help="clamp the 1/(mu-lam) delay branch of the "
+                         "DIFFERENTIABLE path at this value (0 = off): "
+                         "pole mitigation for long-horizon stability "
+                         "(docs/TRAINING.md); the empirical evaluator is "
+                         "never clamped")
+    ap.add_argument("--capture", action="store_true",
+                    help="hipGraph-capture the whole training step "
+                         "(sample→episode→all-reduce→fused Adam) and "
+                         "replay it; explore lives in a device scalar so "
+                         "its decay survives capture; graphs are "
+                         "re-captured on lr changes")
     args = ap.parse_args(argv)
 
     rank, world = dp.init_from_env()
@@ -207,7 +219,8 @@ def main(argv=None):
             cases += [c.pad_to(n_max) for c in build_training_cases(
                 n, per_size, args.distinct, args.T,
                 args.seed + 1000 * rank + 17 * n, workers=args.workers)]
-        engines = [EpisodeEngine(cases, model, device=device, dtype=dtype)]
+        engines = [EpisodeEngine(cases, model, device=device, dtype=dtype,
+                                 delay_clamp=args.delay_clamp)]
     else:
         engines = []
         for n in sizes:
@@ -215,7 +228,8 @@ def main(argv=None):
                 n, per_size, args.distinct, args.T,
                 args.seed + 1000 * rank + 17 * n, workers=args.workers)
             engines.append(EpisodeEngine(cases, model, device=device,
-                                         dtype=dtype))
+                                         dtype=dtype,
+                                         delay_clamp=args.delay_clamp))
     engine = engines[0]
     dp.broadcast_params(engine.model)
     from ..ops import dispatch as mho_dispatch
@@ -233,6 +247,8 @@ def main(argv=None):
         reducer = dp.FlatAllreduce(engine.model.parameters())
     gen = torch.Generator(device=device)
     gen.manual_seed(args.seed * 7919 + rank)
+    for i, e in enumerate(engines):
+        e.set_rng_seed(args.seed * 7919 + 131 * rank + i)
 
     eval_engines = engines
     if args.eval_every and not args.eval_on_train:
@@ -269,20 +285,61 @@ def main(argv=None):
         if device.startswith("cuda"):
             acts.append(ProfilerActivity.CUDA)
         profiler = profile(activities=acts, acc_events=True)
+
+    # ---- hipGraph capture of the whole training step ----------------------
+    # explore lives in a device scalar (the captured kernels read its
+    # CURRENT value), randomness comes from torch's graph-safe default
+    # generator plus the engine's counter RNG, and lr changes invalidate
+    # the captured graphs (the fused-Adam lr is baked at capture time).
+    use_capture = (args.capture and use_fused
+                   and device.startswith("cuda"))
+    explore_dev = None
+    if use_capture:
+        assert world == 1 or True  # all-reduce capture is supported by RCCL
+        explore_dev = torch.tensor(float(args.explore),
+                                   dtype=torch.float32, device=device)
+        torch.manual_seed(args.seed * 7919 + rank)
+        gen = None
+    hip_graphs = {}
+
+    def _fused_step(eng, load, explore_arg):
+        opt.zero_grad()
+        jobs = eng.sample_jobs(load, gen)
+        res = eng.gnn_episode(jobs, explore=explore_arg, gen=gen, train=True)
+        if world > 1:
+            torch.distributed.all_reduce(opt.flat_g)
+        opt.step(scale=1.0 / (eng.B * world))
+        return res
+
+    def _captured_step(eng_idx, load_idx):
+        key = (eng_idx, load_idx)
+        if key not in hip_graphs:
+            eng = engines[eng_idx]
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            _fused_step(eng, loads[load_idx], explore_dev)  # warm side effects
+            with torch.cuda.graph(g):
+                res = _fused_step(eng, loads[load_idx], explore_dev)
+            hip_graphs[key] = (g, res)
+        g, res = hip_graphs[key]
+        g.replay()
+        return res
+
     for step in range(1, args.steps + 1):
         if profiler is not None and not profiler_on and \
                 step >= max(args.steps - args.torch_profile, 1):
             profiler.__enter__()
             profiler_on = True
         engine = engines[step % len(engines)]
-        jobs = engine.sample_jobs(loads[step % len(loads)], gen)
-        if use_fused:
+        if use_capture:
+            res = _captured_step(step % len(engines), step % len(loads))
+            explore_dev.mul_(args.explore_decay).clamp_(min=0.001)
+            pre_norms = None
+        elif use_fused:
+            jobs = engine.sample_jobs(loads[step % len(loads)], gen)
             opt.zero_grad()
-        else:
-            for p in engine.model.parameters():
-                p.grad = None
-        res = engine.gnn_episode(jobs, explore=explore, gen=gen, train=True)
-        if use_fused:
+            res = engine.gnn_episode(jobs, explore=explore, gen=gen,
+                                     train=True)
             scale = 1.0 / (engine.B * world)
             if world > 1:
                 torch.distributed.all_reduce(opt.flat_g)
@@ -291,6 +348,11 @@ def main(argv=None):
                              for p in engine.model.parameters()]
             opt.step(scale=scale)
         else:
+            jobs = engine.sample_jobs(loads[step % len(loads)], gen)
+            for p in engine.model.parameters():
+                p.grad = None
+            res = engine.gnn_episode(jobs, explore=explore, gen=gen,
+                                     train=True)
             with torch.no_grad():
                 for p in engine.model.parameters():
                     if p.grad is not None:
@@ -310,6 +372,7 @@ def main(argv=None):
         if args.lr_decay_at and step == args.lr_decay_at:
             for group in opt.param_groups:
                 group["lr"] *= 0.1
+            hip_graphs.clear()      # lr is baked into captured graphs
         if args.eval_every and step % args.eval_every == 0:
             eval_tau = _dist_mean(
                 evaluate_policy(eval_engines, loads, args.eval_seed,
@@ -340,6 +403,7 @@ def main(argv=None):
                 opt.load_state_dict(best_opt)
                 for group in opt.param_groups:
                     group["lr"] /= 3.0
+                hip_graphs.clear()  # lr is baked into captured graphs
                 rollbacks += 1
                 if rank == 0:
                     print(json.dumps({"step": step, "rollback": rollbacks,
@@ -349,6 +413,11 @@ def main(argv=None):
                           flush=True)
 
         if step % args.log_every == 0 and rank == 0:
+            if pre_norms is None:   # capture mode: flat_g still holds the
+                # replayed step's summed gradients
+                with torch.no_grad():
+                    pre_norms = [float(p.grad.norm()) / (engine.B * world)
+                                 for p in engine.model.parameters()]
             tau = float(torch.nanmean(res.tau))
             congest = int(res.congest.sum())
             njobs = int(res.num_jobs.sum())

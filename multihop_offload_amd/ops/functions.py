@@ -8,13 +8,14 @@ from . import dispatch
 
 
 def cheb_compat(model) -> bool:
-    """The fused ChebConv kernel covers the reference architecture family:
-    K<=2, all feature widths <=32, no dropout."""
+    """The fused ChebConv kernels cover the reference architecture family:
+    any K (K<=2 on the tuned flagship kernels, K>=3 on the generic-K
+    recurrence kernels), all feature widths <=32, no dropout."""
     if getattr(model, "dropout", 0.0) > 0:
         return False
     for layer in model.layers:
         K, fi, fo = layer.weight.shape
-        if K > 2 or fi > 32 or fo > 32:
+        if fi > 32 or fo > 32:
             return False
     return True
 
@@ -41,8 +42,21 @@ class ChebStackFn(torch.autograd.Function):
             w, b = params[2 * l], params[2 * l + 1]
             Wp[l, :w.shape[0], :w.shape[1], :w.shape[2]] = w
             bp[l, :b.shape[0]] = b
+        ctx.K = K
         ctx.large = not cheb_lds_fits(eng.Ee, K)
-        if ctx.large:
+        if K > 2:
+            # generic-K recurrence kernels (3 LDS row buffers); no
+            # large-graph variant — refuse loudly rather than fall back
+            if ctx.large:
+                raise RuntimeError(
+                    f"fused ChebConv with K={K} does not fit LDS at "
+                    f"Ee={eng.Ee}; no GPU path for this configuration "
+                    "(set MHO_ALLOW_TORCH_GPU=1 to accept eager torch)")
+            lam, acts, t1s = ext.cheb_kn_fwd(x, Wp, bp, eng.k_ext_indptr,
+                                             eng.k_ext_base, eng.k_ext_cols,
+                                             eng.k_ext_max_nnz)
+            ctx.t1s = t1s
+        elif ctx.large:
             lam, acts = ext.cheb_large_fwd(x, Wp, bp, eng.k_ext_indptr,
                                            eng.k_ext_base, eng.k_ext_cols)
         else:
@@ -60,7 +74,11 @@ class ChebStackFn(torch.autograd.Function):
         acts, Wp = ctx.saved_tensors
         eng = ctx.eng
         ext = dispatch.require_hip()
-        if ctx.large:
+        if ctx.K > 2:
+            dW, db = ext.cheb_kn_bwd(dlam.contiguous(), acts, ctx.t1s, Wp,
+                                     eng.k_ext_indptr, eng.k_ext_base,
+                                     eng.k_ext_cols, eng.k_ext_max_nnz)
+        elif ctx.large:
             dW, db = ext.cheb_large_bwd(dlam.contiguous(), acts, Wp,
                                         eng.k_ext_indptr, eng.k_ext_base,
                                         eng.k_ext_cols)
@@ -95,7 +113,8 @@ class ActorHeadFn(torch.autograd.Function):
             lam_ext.contiguous(), eng.k_conf_indptr, eng.k_conf_base,
             eng.k_conf_cols, eng.link_rates.contiguous(),
             eng.bw_comp.contiguous(), eng.k_edges, eng.node_vedge,
-            eng.T_arr.contiguous(), eng.k_E_arr, eng.N, eng.fp_iters)
+            eng.T_arr.contiguous(), eng.k_E_arr, eng.N, eng.fp_iters,
+            eng.delay_clamp)
         ctx.save_for_backward(lam_ext, mu_hist)
         ctx.eng = eng
         return dm
@@ -110,7 +129,7 @@ class ActorHeadFn(torch.autograd.Function):
             eng.k_conf_indptr, eng.k_conf_base, eng.k_conf_cols,
             eng.link_rates.contiguous(), eng.bw_comp.contiguous(),
             eng.k_edges, eng.node_vedge, eng.T_arr.contiguous(),
-            eng.k_E_arr, eng.fp_iters)
+            eng.k_E_arr, eng.fp_iters, eng.delay_clamp)
         return dlam, None
 
 

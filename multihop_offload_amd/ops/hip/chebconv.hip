@@ -208,11 +208,12 @@ DEV_INLINE void load_acts(float* __restrict__ lds,
 }
 
 // ---- SpMV over the support: dst[r][:] (+)= sum_nb src[nb][:] -------------
-// sign=+1: dst = A·src (overwrite); inplace2: dst[r] = 2*(A·src)[r]-dst[r]
+// modes: 0: dst = A·src; 1: dst += A·src; 2 (Chebyshev in place):
+// dst = 2*(A·src) - dst; 3 (reverse recurrence): dst += 2*(A·src)
 DEV_INLINE void spmv(const float* __restrict__ src, float* __restrict__ dst,
                      const int* __restrict__ indptr,
                      const int* __restrict__ cols, int Ee, int rows_pad,
-                     int tid, int nt, int mode /*0: =, 1: +=, 2: cheb*/) {
+                     int tid, int nt, int mode) {
     // thread handles (row, 8-col group)
     for (int task = tid; task < rows_pad * (F / 8); task += nt) {
         const int r = task / (F / 8);
@@ -232,10 +233,48 @@ DEV_INLINE void spmv(const float* __restrict__ src, float* __restrict__ dst,
         } else if (mode == 1) {
 #pragma unroll
             for (int c = 0; c < 8; ++c) d[c] += acc[c];
-        } else {
+        } else if (mode == 2) {
 #pragma unroll
             for (int c = 0; c < 8; ++c) d[c] = 2.f * acc[c] - d[c];
+        } else {
+#pragma unroll
+            for (int c = 0; c < 8; ++c) d[c] += 2.f * acc[c];
         }
+    }
+}
+
+// ---- small LDS-tile helpers for the generic-K path -----------------------
+DEV_INLINE void zero_rows(float* __restrict__ buf, int rows_pad, int tid,
+                          int nt) {
+    for (int t = tid; t < rows_pad * F; t += nt)
+        buf[(t / F) * STRIDE + (t % F)] = 0.f;
+}
+
+DEV_INLINE void negate_rows(float* __restrict__ buf, int rows_pad, int tid,
+                            int nt) {
+    for (int t = tid; t < rows_pad * F; t += nt) {
+        float* p = buf + (t / F) * STRIDE + (t % F);
+        *p = -*p;
+    }
+}
+
+DEV_INLINE void add_rows(float* __restrict__ dst,
+                         const float* __restrict__ src, int rows_pad,
+                         int tid, int nt) {
+    for (int t = tid; t < rows_pad * F; t += nt)
+        dst[(t / F) * STRIDE + (t % F)] += src[(t / F) * STRIDE + (t % F)];
+}
+
+// dst = act(src + bias): leaky_relu(0.2) hidden, relu on the last layer
+DEV_INLINE void bias_act_rows(const float* __restrict__ src,
+                              float* __restrict__ dst,
+                              const float* __restrict__ bl, bool last,
+                              int rows_pad, int tid, int nt) {
+    for (int t = tid; t < rows_pad * F; t += nt) {
+        const int r = t / F, c = t % F;
+        float y = src[r * STRIDE + c] + bl[c];
+        y = last ? (y > 0.f ? y : 0.f) : (y > 0.f ? y : 0.2f * y);
+        dst[r * STRIDE + c] = y;
     }
 }
 
@@ -418,6 +457,213 @@ __global__ void cheb_bwd_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Generic-K (K >= 2, covers K > 2) fused ChebConv stack.  Three LDS row
+// buffers carry the Chebyshev recurrence T_k = 2·A·T_{k-1} − T_{k-2}
+// (T_0 = X, T_1 = A·X; models/chebconv.py:54-63): Xb holds T_{k-2} and is
+// consumed in place, Tb holds T_{k-1}, Yb accumulates Σ_k T_k·W_k.  The
+// per-layer T_k (k >= 1) land in `tks` for the backward's weight
+// gradients; the backward reverses the recurrence with the adjoint
+// ĝ_{k-1} += 2·A·ĝ_k, ĝ_{k-2} −= ĝ_k (A symmetric), dX = ĝ_0 + A·ĝ_1.
+// The K<=2 kernels above stay the tuned flagship path.
+// ---------------------------------------------------------------------------
+__global__ void cheb_kn_fwd_kernel(
+    const float* __restrict__ x_in,      // (B,Ee,4)
+    const float* __restrict__ W,         // (L,K,32,32) padded
+    const float* __restrict__ bias,      // (L,32)
+    const int* __restrict__ ext_indptr,  // (B,Ee+1)
+    const long* __restrict__ ext_base,   // (B)
+    const int* __restrict__ ext_cols,
+    float* __restrict__ acts,            // (B,L+1,Ee,32) out
+    float* __restrict__ tks,             // (B,L,K-1,Ee,32) out
+    float* __restrict__ lam,             // (B,Ee) out
+    int B, int Ee, int L, int K, int rows_pad, int stage_csr) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    float* Xb = reinterpret_cast<float*>(smem_raw);
+    float* Tb = Xb + (size_t)rows_pad * STRIDE;
+    float* Yb = Tb + (size_t)rows_pad * STRIDE;
+    float* Wl = Yb + (size_t)rows_pad * STRIDE;   // K*32*32
+    float* bl = Wl + (size_t)K * F * F;           // 32
+
+    const int b = blockIdx.x;
+    const int tid = threadIdx.x, nt = blockDim.x;
+    const int* ipt = ext_indptr + (size_t)b * (Ee + 1);
+    const int* cls = ext_cols + ext_base[b];
+    if (stage_csr) {
+        int* l_ipt = reinterpret_cast<int*>(bl + F);
+        int* l_cols = l_ipt + (Ee + 1);
+        for (int i = tid; i < Ee + 1; i += nt) l_ipt[i] = ipt[i];
+        const int nnz = ipt[Ee];
+        for (int i = tid; i < nnz; i += nt) l_cols[i] = cls[i];
+        ipt = l_ipt;
+        cls = l_cols;
+    }
+    const float* xb = x_in + (size_t)b * Ee * 4;
+    float* actsb = acts + (size_t)b * (L + 1) * Ee * F;
+    float* tksb = tks + (size_t)b * L * (K - 1) * Ee * F;
+
+    for (int r = tid; r < rows_pad; r += nt) {
+        float* row = Xb + r * STRIDE;
+        for (int c = 0; c < F; ++c) row[c] = 0.f;
+        if (r < Ee) {
+            const float4 v =
+                *reinterpret_cast<const float4*>(xb + (size_t)r * 4);
+            row[0] = v.x; row[1] = v.y; row[2] = v.z; row[3] = v.w;
+        }
+    }
+    __syncthreads();
+    store_acts(Xb, actsb, Ee, tid, nt);
+
+    for (int l = 0; l < L; ++l) {
+        for (int i = tid; i < K * F * F; i += nt)
+            Wl[i] = W[((size_t)l * K) * F * F + i];
+        for (int i = tid; i < F; i += nt) bl[i] = bias[l * F + i];
+        zero_rows(Yb, rows_pad, tid, nt);
+        __syncthreads();
+        gemm_acc(Xb, Yb, Wl, false, rows_pad, tid);           // T0·W0
+        spmv(Xb, Tb, ipt, cls, Ee, rows_pad, tid, nt, 0);     // T1 = A·X
+        __syncthreads();
+        store_acts(Tb, tksb + (size_t)l * (K - 1) * Ee * F, Ee, tid, nt);
+        gemm_acc(Tb, Yb, Wl + F * F, false, rows_pad, tid);   // T1·W1
+        float* prev = Xb;                                     // T_{k-2}
+        float* cur = Tb;                                      // T_{k-1}
+        for (int k = 2; k < K; ++k) {
+            __syncthreads();
+            spmv(cur, prev, ipt, cls, Ee, rows_pad, tid, nt, 2);
+            __syncthreads();
+            float* t = prev; prev = cur; cur = t;             // cur = T_k
+            store_acts(cur, tksb + ((size_t)l * (K - 1) + k - 1) * Ee * F,
+                       Ee, tid, nt);
+            gemm_acc(cur, Yb, Wl + (size_t)k * F * F, false, rows_pad, tid);
+        }
+        __syncthreads();
+        bias_act_rows(Yb, Xb, bl, l == L - 1, rows_pad, tid, nt);
+        __syncthreads();
+        store_acts(Xb, actsb + (size_t)(l + 1) * Ee * F, Ee, tid, nt);
+        __syncthreads();
+    }
+    for (int r = tid; r < Ee; r += nt) lam[(size_t)b * Ee + r] =
+        Xb[r * STRIDE];
+}
+
+__global__ void cheb_kn_bwd_kernel(
+    const float* __restrict__ dlam,      // (B,Ee)
+    const float* __restrict__ acts,      // (B,L+1,Ee,32)
+    const float* __restrict__ tks,       // (B,L,K-1,Ee,32)
+    const float* __restrict__ W,         // (L,K,32,32)
+    const int* __restrict__ ext_indptr,
+    const long* __restrict__ ext_base,
+    const int* __restrict__ ext_cols,
+    float* __restrict__ dW,              // (B,L,K,32,32) out (prezeroed)
+    float* __restrict__ db,              // (B,L,32) out (prezeroed)
+    int B, int Ee, int L, int K, int rows_pad, int stage_csr) {
+    extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+    float* Db = reinterpret_cast<float*>(smem_raw);   // delta
+    float* Pb = Db + (size_t)rows_pad * STRIDE;
+    float* Qb = Pb + (size_t)rows_pad * STRIDE;
+    float* Wl = Qb + (size_t)rows_pad * STRIDE;       // K*F*F
+
+    const int b = blockIdx.x;
+    const int tid = threadIdx.x, nt = blockDim.x;
+    const int* ipt = ext_indptr + (size_t)b * (Ee + 1);
+    const int* cls = ext_cols + ext_base[b];
+    if (stage_csr) {
+        int* l_ipt = reinterpret_cast<int*>(Wl + (size_t)K * F * F);
+        int* l_cols = l_ipt + (Ee + 1);
+        for (int i = tid; i < Ee + 1; i += nt) l_ipt[i] = ipt[i];
+        const int nnz = ipt[Ee];
+        for (int i = tid; i < nnz; i += nt) l_cols[i] = cls[i];
+        ipt = l_ipt;
+        cls = l_cols;
+    }
+    const float* actsb = acts + (size_t)b * (L + 1) * Ee * F;
+    const float* tksb = tks + (size_t)b * L * (K - 1) * Ee * F;
+    float* dWb = dW + (size_t)b * L * K * F * F;
+    float* dbb = db + (size_t)b * L * F;
+
+    for (int r = tid; r < rows_pad; r += nt) {
+        float* row = Db + r * STRIDE;
+        for (int c = 0; c < F; ++c) row[c] = 0.f;
+        if (r < Ee) row[0] = dlam[(size_t)b * Ee + r];
+    }
+    __syncthreads();
+
+    for (int l = L - 1; l >= 0; --l) {
+        const bool last = (l == L - 1);
+        const float slope = last ? 0.f : 0.2f;
+        for (int t = tid; t < Ee * (F / 4); t += nt) {
+            const int r = t >> 3;
+            const int c = (t & 7) * 4;
+            const float4 v = *reinterpret_cast<const float4*>(
+                actsb + ((size_t)(l + 1) * Ee + r) * F + c);
+            float* d = Db + r * STRIDE + c;
+            d[0] *= v.x > 0.f ? 1.f : slope;
+            d[1] *= v.y > 0.f ? 1.f : slope;
+            d[2] *= v.z > 0.f ? 1.f : slope;
+            d[3] *= v.w > 0.f ? 1.f : slope;
+        }
+        for (int i = tid; i < K * F * F; i += nt)
+            Wl[i] = W[((size_t)l * K) * F * F + i];
+        __syncthreads();
+        {   // db[j] = sum_r Db[r][j]
+            const int nchunk = nt / F;
+            const int j = tid % F, ch = tid / F;
+            float acc = 0.f;
+            for (int r = ch; r < Ee; r += nchunk) acc += Db[r * STRIDE + j];
+            atomicAdd(&dbb[l * F + j], acc);
+        }
+        // weight gradients: dW_k = T_kᵀ·Db, T_k streamed through Pb
+        load_acts(Pb, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
+        __syncthreads();
+        gemm_wgrad(Pb, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
+        for (int k = 1; k < K; ++k) {
+            __syncthreads();
+            load_acts(Pb, tksb + ((size_t)l * (K - 1) + k - 1) * Ee * F,
+                      Ee, rows_pad, tid, nt);
+            __syncthreads();
+            gemm_wgrad(Pb, Db, dWb + ((size_t)l * K + k) * F * F, rows_pad,
+                       tid);
+        }
+        if (l == 0) break;
+        __syncthreads();
+        // dX via the reverse recurrence (buffers: Pb=ĝ_cur, Qb=ĝ_prev)
+        zero_rows(Pb, rows_pad, tid, nt);
+        zero_rows(Qb, rows_pad, tid, nt);
+        __syncthreads();
+        if (K == 1) {
+            gemm_acc(Db, Pb, Wl, true, rows_pad, tid);        // ĝ_0
+            __syncthreads();
+            // dX = ĝ_0 → copy into Db
+            for (int t = tid; t < rows_pad * F; t += nt)
+                Db[(t / F) * STRIDE + (t % F)] =
+                    Pb[(t / F) * STRIDE + (t % F)];
+        } else {
+            gemm_acc(Db, Pb, Wl + (size_t)(K - 1) * F * F, true, rows_pad,
+                     tid);                                    // ĝ_{K-1}
+            gemm_acc(Db, Qb, Wl + (size_t)(K - 2) * F * F, true, rows_pad,
+                     tid);                                    // ĝ_{K-2} base
+            float* cur = Pb;
+            float* prv = Qb;
+            for (int k = K - 1; k >= 2; --k) {
+                __syncthreads();
+                spmv(cur, prv, ipt, cls, Ee, rows_pad, tid, nt, 3);
+                __syncthreads();
+                negate_rows(cur, rows_pad, tid, nt);          // −ĝ_k
+                __syncthreads();
+                gemm_acc(Db, cur, Wl + (size_t)(k - 2) * F * F, true,
+                         rows_pad, tid);                      // + base_{k-2}
+                float* t = cur; cur = prv; prv = t;           // cur=ĝ_{k-1}
+            }
+            __syncthreads();
+            // dX = ĝ_0 + A·ĝ_1  (cur=ĝ_1, prv=ĝ_0)
+            spmv(cur, Db, ipt, cls, Ee, rows_pad, tid, nt, 0);
+            __syncthreads();
+            add_rows(Db, prv, rows_pad, tid, nt);
+        }
+        __syncthreads();
+    }
+}
+
 }  // namespace
 
 static int round16(int x) { return (x + 15) & ~15; }
@@ -449,6 +695,64 @@ std::vector<torch::Tensor> cheb_fwd_hip(
                        lam.data_ptr<float>(),
                        B, Ee, L, K, rows_pad, (int)max_nnz, stage_csr);
     return {lam, acts, t1s};
+}
+
+std::vector<torch::Tensor> cheb_kn_fwd_hip(
+    torch::Tensor x, torch::Tensor W, torch::Tensor bias,
+    torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor ext_cols, long max_nnz) {
+    const int B = x.size(0), Ee = x.size(1);
+    const int L = W.size(0), K = W.size(1);
+    TORCH_CHECK(K >= 2, "generic-K ChebConv kernel expects K>=2");
+    const int rows_pad = round16(Ee);
+    auto acts = torch::empty({B, L + 1, Ee, F}, x.options());
+    auto tks = torch::empty({B, L, K - 1, Ee, F}, x.options());
+    auto lam = torch::empty({B, Ee}, x.options());
+    size_t lds = sizeof(float) *
+        (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F + F);
+    TORCH_CHECK(lds <= 160 * 1024,
+                "graph too large for generic-K fused ChebConv (LDS)");
+    const size_t csr_bytes = sizeof(int) * ((size_t)Ee + 1 + max_nnz);
+    const int stage_csr = (lds + csr_bytes <= 160 * 1024) ? 1 : 0;
+    if (stage_csr) lds += csr_bytes;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(cheb_kn_fwd_kernel, dim3(B), dim3(512), lds,
+                       stream.stream(),
+                       x.data_ptr<float>(), W.data_ptr<float>(),
+                       bias.data_ptr<float>(), ext_indptr.data_ptr<int>(),
+                       ext_base.data_ptr<long>(), ext_cols.data_ptr<int>(),
+                       acts.data_ptr<float>(), tks.data_ptr<float>(),
+                       lam.data_ptr<float>(),
+                       B, Ee, L, K, rows_pad, stage_csr);
+    return {lam, acts, tks};
+}
+
+std::vector<torch::Tensor> cheb_kn_bwd_hip(
+    torch::Tensor dlam, torch::Tensor acts, torch::Tensor tks,
+    torch::Tensor W, torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor ext_cols, long max_nnz) {
+    const int B = dlam.size(0), Ee = dlam.size(1);
+    const int L = W.size(0), K = W.size(1);
+    const int rows_pad = round16(Ee);
+    auto dW = torch::zeros({B, L, K, F, F}, dlam.options());
+    auto db = torch::zeros({B, L, F}, dlam.options());
+    size_t lds = sizeof(float) *
+        (3 * (size_t)rows_pad * STRIDE + (size_t)K * F * F);
+    TORCH_CHECK(lds <= 160 * 1024,
+                "graph too large for generic-K fused ChebConv bwd (LDS)");
+    const size_t csr_bytes = sizeof(int) * ((size_t)Ee + 1 + max_nnz);
+    const int stage_csr = (lds + csr_bytes <= 160 * 1024) ? 1 : 0;
+    if (stage_csr) lds += csr_bytes;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(cheb_kn_bwd_kernel, dim3(B), dim3(512), lds,
+                       stream.stream(),
+                       dlam.data_ptr<float>(), acts.data_ptr<float>(),
+                       tks.data_ptr<float>(), W.data_ptr<float>(),
+                       ext_indptr.data_ptr<int>(), ext_base.data_ptr<long>(),
+                       ext_cols.data_ptr<int>(),
+                       dW.data_ptr<float>(), db.data_ptr<float>(),
+                       B, Ee, L, K, rows_pad, stage_csr);
+    return {dW, db};
 }
 
 std::vector<torch::Tensor> cheb_bwd_hip(

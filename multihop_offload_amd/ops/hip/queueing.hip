@@ -82,18 +82,24 @@ DEV_INLINE void fixed_point_bwd(const float* lam, const float* rates,
     }
 }
 
-// unit = 1/(mu-lam), overwritten by T*lam/(denom*mu) where lam-mu > 0
-DEV_INLINE float unit_fwd(float lam, float mu, float T, float denom) {
-    return (lam - mu) > 0.f ? T * lam / (denom * mu) : 1.0f / (mu - lam);
+// unit = 1/(mu-lam), overwritten by T*lam/(denom*mu) where lam-mu > 0.
+// cap > 0 clamps the 1/(mu-lam) branch (torch.clamp semantics: value
+// capped, gradient zero where inv > cap) — pole mitigation for training.
+DEV_INLINE float unit_fwd(float lam, float mu, float T, float denom,
+                          float cap) {
+    if ((lam - mu) > 0.f) return T * lam / (denom * mu);
+    const float inv = 1.0f / (mu - lam);
+    return (cap > 0.f && inv > cap) ? cap : inv;
 }
 // cotangents: given dunit, produce (dlam, dmu) contributions
 DEV_INLINE void unit_bwd(float lam, float mu, float T, float denom,
-                         float dunit, float* dlam, float* dmu) {
+                         float cap, float dunit, float* dlam, float* dmu) {
     if ((lam - mu) > 0.f) {
         *dlam = dunit * T / (denom * mu);
         *dmu = -dunit * T * lam / (denom * mu * mu);
     } else {
         const float inv = 1.0f / (mu - lam);
+        if (cap > 0.f && inv > cap) { *dlam = 0.f; *dmu = 0.f; return; }
         *dlam = dunit * inv * inv;
         *dmu = -dunit * inv * inv;
     }
@@ -124,6 +130,7 @@ __global__ void critic_kernel(
     float* __restrict__ g_dunit,           // (B,Ee) scratch (prezeroed)
     float* __restrict__ g_dlam,            // (B,Ee) scratch (prezeroed)
     int large,                             // 1: use global scratch
+    float cap,                             // delay clamp (0 = off)
     int E, int C, int Ee, int J, int H, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     const int b_ = blockIdx.x;
@@ -189,9 +196,9 @@ __global__ void critic_kernel(
     fixed_point_fwd(lam_e, ratesb, cip, ccols, hist, s1, E, iters, tid, nt);
     const float* mu_last = hist + (size_t)iters * E;
     for (int e = tid; e < E; e += nt)
-        unit[e] = unit_fwd(lam_e[e], mu_last[e], T, 101.0f);
+        unit[e] = unit_fwd(lam_e[e], mu_last[e], T, 101.0f, cap);
     for (int k = tid; k < C; k += nt)
-        unit[E + k] = unit_fwd(lam_e[E + k], bwb[k], T, 100.0f);
+        unit[E + k] = unit_fwd(lam_e[E + k], bwb[k], T, 100.0f, cap);
     __syncthreads();
 
     // ---- loss + dL/dunit over route entries ------------------------------
@@ -218,13 +225,13 @@ __global__ void critic_kernel(
     // ---- reverse: dunit → (dlam over links via fixed point, direct nodes)
     for (int e = tid; e < E; e += nt) {
         float dl_, dm_;
-        unit_bwd(lam_e[e], mu_last[e], T, 101.0f, dunit[e], &dl_, &dm_);
+        unit_bwd(lam_e[e], mu_last[e], T, 101.0f, cap, dunit[e], &dl_, &dm_);
         dlam[e] += dl_;
         s2[e] = dm_;                               // dmu into the reverse
     }
     for (int k = tid; k < C; k += nt) {
         float dl_, dm_;
-        unit_bwd(lam_e[E + k], bwb[k], T, 100.0f, dunit[E + k], &dl_, &dm_);
+        unit_bwd(lam_e[E + k], bwb[k], T, 100.0f, cap, dunit[E + k], &dl_, &dm_);
         dlam[E + k] += dl_;                        // bw is constant
     }
     __syncthreads();
@@ -277,7 +284,7 @@ __global__ void actor_head_fwd_kernel(
     float* __restrict__ mu_hist_out,       // (B,(iters+1),E) out
     const float* __restrict__ T_arr,
     const int* __restrict__ E_arr,         // (B) real link counts (ragged)
-    int large, int N, int E, int C, int Ee, int iters) {
+    int large, float cap, int N, int E, int C, int Ee, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);       // E
     float* hist;
@@ -308,7 +315,7 @@ __global__ void actor_head_fwd_kernel(
     const float* mu_last = hist + (size_t)iters * E;
     const int Eb = E_arr[b];
     for (int e = tid; e < Eb; e += nt) {
-        const float d = unit_fwd(lam[e], mu_last[e], T, 101.0f);
+        const float d = unit_fwd(lam[e], mu_last[e], T, 101.0f, cap);
         const int u = edg[e * 2], v = edg[e * 2 + 1];
         dmb[(size_t)u * N + v] = d;
         dmb[(size_t)v * N + u] = d;
@@ -316,7 +323,7 @@ __global__ void actor_head_fwd_kernel(
     for (int n = tid; n < N; n += nt) {
         const long ve = nv[n];
         dmb[(size_t)n * N + n] =
-            ve >= 0 ? unit_fwd(le[ve], bwb[ve - E], T, 100.0f) : INFINITY;
+            ve >= 0 ? unit_fwd(le[ve], bwb[ve - E], T, 100.0f, cap) : INFINITY;
     }
     if (!large) {
         float* ho = mu_hist_out + (size_t)b * (iters + 1) * E;
@@ -341,7 +348,7 @@ __global__ void actor_head_bwd_kernel(
     float* __restrict__ dlam_ext,          // (B,Ee) out
     const float* __restrict__ T_arr,
     const int* __restrict__ E_arr,
-    int large, int N, int E, int C, int Ee, int iters) {
+    int large, float cap, int N, int E, int C, int Ee, int iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);
     const float* hist;
@@ -390,7 +397,7 @@ __global__ void actor_head_bwd_kernel(
             dd = gd[(size_t)u * N + v] + gd[(size_t)v * N + u];
         }
         float dl_, dm_;
-        unit_bwd(lam[e], mu_last[e], T, 101.0f, dd, &dl_, &dm_);
+        unit_bwd(lam[e], mu_last[e], T, 101.0f, cap, dd, &dl_, &dm_);
         dlam[e] = dl_;
         dmu[e] = dm_;
     }
@@ -404,8 +411,8 @@ __global__ void actor_head_bwd_kernel(
         const long ve = nv[n];
         if (ve >= 0) {
             float dl_, dm_;
-            unit_bwd(le[ve], bwb[ve - E], T, 100.0f, gd[(size_t)n * N + n],
-                     &dl_, &dm_);
+            unit_bwd(le[ve], bwb[ve - E], T, 100.0f, cap,
+                     gd[(size_t)n * N + n], &dl_, &dm_);
             out[ve] = dl_;
         }
     }
@@ -422,7 +429,7 @@ std::vector<torch::Tensor> critic_hip(
     torch::Tensor mask, torch::Tensor rate, torch::Tensor ul,
     torch::Tensor dl, torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
-    torch::Tensor T_arr, long Ee, long iters) {
+    torch::Tensor T_arr, long Ee, long iters, double cap) {
     const int B = route_links.size(0), J = route_links.size(1);
     const int H = route_links.size(2);
     const int E = rates.size(1), C = bw_comp.size(1);
@@ -459,7 +466,7 @@ std::vector<torch::Tensor> critic_hip(
                        grad_edge.data_ptr<float>(), loss.data_ptr<float>(),
                        T_arr.data_ptr<float>(),
                        g_hist.data_ptr<float>(), g_dunit.data_ptr<float>(),
-                       g_dlam.data_ptr<float>(), large,
+                       g_dlam.data_ptr<float>(), large, (float)cap,
                        E, C, (int)Ee, J, H, (int)iters);
     return {grad_edge, loss};
 }
@@ -468,7 +475,8 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
     torch::Tensor lam_ext, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
     torch::Tensor bw_comp, torch::Tensor edges, torch::Tensor node_vedge,
-    torch::Tensor T_arr, torch::Tensor E_arr, long N, long iters) {
+    torch::Tensor T_arr, torch::Tensor E_arr, long N, long iters,
+    double cap) {
     const int B = lam_ext.size(0), Ee = lam_ext.size(1);
     const int E = rates.size(1), C = bw_comp.size(1);
     auto dm = torch::zeros({B, N, N}, lam_ext.options());
@@ -491,7 +499,7 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
                        node_vedge.data_ptr<long>(), dm.data_ptr<float>(),
                        mu_hist.data_ptr<float>(),
                        T_arr.data_ptr<float>(), E_arr.data_ptr<int>(),
-                       large, (int)N, E, C, Ee, (int)iters);
+                       large, (float)cap, (int)N, E, C, Ee, (int)iters);
     return {dm, mu_hist};
 }
 
@@ -500,7 +508,7 @@ torch::Tensor actor_head_bwd_hip(
     torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
     torch::Tensor edges, torch::Tensor node_vedge, torch::Tensor T_arr,
-    torch::Tensor E_arr, long iters) {
+    torch::Tensor E_arr, long iters, double cap) {
     const int B = lam_ext.size(0), Ee = lam_ext.size(1);
     const int E = rates.size(1), C = bw_comp.size(1);
     const int N = grad_dist.size(1);
@@ -523,6 +531,6 @@ torch::Tensor actor_head_bwd_hip(
                        bw_comp.data_ptr<float>(), edges.data_ptr<int>(),
                        node_vedge.data_ptr<long>(), dlam.data_ptr<float>(),
                        T_arr.data_ptr<float>(), E_arr.data_ptr<int>(),
-                       large, N, E, C, Ee, (int)iters);
+                       large, (float)cap, N, E, C, Ee, (int)iters);
     return dlam;
 }

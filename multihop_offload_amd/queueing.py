@@ -64,17 +64,26 @@ def fixed_point_mu(lam: torch.Tensor, rates: torch.Tensor, cf_degs: torch.Tensor
 
 
 def delay_with_fallback(lam: torch.Tensor, mu: torch.Tensor, T,
-                        denom: float) -> torch.Tensor:
+                        denom: float, cap: float = 0.0) -> torch.Tensor:
     """delay = 1/(mu-lam), replaced by T*lam/(denom*mu) where lam > mu.
     ``T`` may be a scalar or a tensor broadcastable against ``lam``
     (per-graph horizons).
 
     Matches the TF semantics: the congested entries are *overwritten* by the
     fallback (tensor_scatter_nd_update), so no gradient flows through the
-    1/(mu-lam) branch at those entries."""
+    1/(mu-lam) branch at those entries.
+
+    ``cap`` > 0 clamps the 1/(mu-lam) branch at ``cap`` (torch.clamp
+    semantics: zero gradient where clamped).  This is the pole mitigation
+    for training stability — as lam→mu⁻ the delay and its gradient
+    1/(mu-lam)² blow up, which is the diagnosed driver of late-training
+    collapse (docs/TRAINING.md).  The congested fallback branch (already
+    bounded) is never clamped; 0 disables (reference semantics)."""
     congested = (lam - mu) > 0
     safe = torch.where(congested, torch.ones_like(mu), mu - lam)
     normal = 1.0 / safe
+    if cap > 0:
+        normal = normal.clamp(max=cap)
     fallback = T * lam / (denom * mu)
     return torch.where(congested, fallback, normal)
 

@@ -481,3 +481,246 @@ def test_fused_adam_state_dict_roundtrip():
     torch.cuda.synchronize()
     torch.testing.assert_close(after_rollback, o2.flat_p,
                                rtol=1e-6, atol=1e-7)
+
+
+@needs_gpu
+def test_cheb_k3_gpu_matches_cpu():
+    """Generic-K fused ChebConv (K=3): GPU gradients vs the CPU torch
+    recurrence (models/chebconv.py:54-63)."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.graphs import JobInstance
+    from tests.test_engine import _jobbatch_from
+
+    cases = _cases()
+    model_c = ChebConvStack(K=3, dtype=torch.float32, seed=7)
+    model_g = ChebConvStack(K=3, dtype=torch.float32, seed=7)
+    with torch.no_grad():
+        for pc, pg in zip(model_c.parameters(), model_g.parameters()):
+            pc.mul_(0.01)
+            pg.copy_(pc)
+        model_c.layers[-1].bias.fill_(0.5)
+        model_g.layers[-1].bias.fill_(0.5)
+    eng_c = EpisodeEngine(cases, model_c, device="cpu", dtype=torch.float32)
+    eng_g = EpisodeEngine(cases, model_g, device="cuda", dtype=torch.float32)
+    insts = [JobInstance.sample(c.mobile_nodes, 0.15,
+                                np.random.RandomState(40 + i))
+             for i, c in enumerate(cases)]
+    res_c = eng_c.gnn_episode(_jobbatch_from(eng_c, insts), train=True)
+    res_g = eng_g.gnn_episode(_jobbatch_from(eng_g, insts), train=True)
+    assert np.allclose(res_c.tau.numpy(), res_g.tau.cpu().numpy(), rtol=1e-3)
+    for pc, pg in zip(model_c.parameters(), model_g.parameters()):
+        a, b = pc.grad.numpy(), pg.grad.cpu().numpy()
+        denom = max(np.abs(a).max(), 1e-6)
+        assert np.abs(a - b).max() / denom < 5e-3, (a - b)
+
+
+@needs_gpu
+def test_cheb_k4_forward_matches_cpu():
+    """K=4 exercises the in-place Chebyshev ping-pong over >1 recurrence
+    step (forward only — the shipped configs are K<=3)."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.graphs import JobInstance
+    from tests.test_engine import _jobbatch_from
+    cases = _cases(B=4)
+    mc = ChebConvStack(K=4, dtype=torch.float32, seed=9)
+    mg = ChebConvStack(K=4, dtype=torch.float32, seed=9)
+    with torch.no_grad():
+        for pc, pg in zip(mc.parameters(), mg.parameters()):
+            pc.mul_(0.01)
+            pg.copy_(pc)
+    eng_c = EpisodeEngine(cases, mc, device="cpu", dtype=torch.float32)
+    eng_g = EpisodeEngine(cases, mg, device="cuda", dtype=torch.float32)
+    insts = [JobInstance.sample(c.mobile_nodes, 0.15,
+                                np.random.RandomState(60 + i))
+             for i, c in enumerate(cases)]
+    res_c = eng_c.gnn_episode(_jobbatch_from(eng_c, insts), train=False)
+    res_g = eng_g.gnn_episode(_jobbatch_from(eng_g, insts), train=False)
+    assert np.allclose(res_c.tau.numpy(), res_g.tau.cpu().numpy(), rtol=1e-3)
+
+
+@needs_gpu
+def test_in_kernel_explore_uniform():
+    """explore=1.0 in the decide kernel → destination uniform over the
+    valid servers + local (statistical, counter-RNG)."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    cases = _cases(B=8)
+    model = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    eng = EpisodeEngine(cases, model, device="cuda", dtype=torch.float32)
+    eng.set_rng_seed(999)
+    gen = torch.Generator(device="cuda")
+    gen.manual_seed(0)
+    counts = {}
+    total = 0
+    for _ in range(40):
+        jobs = eng.sample_jobs(0.15, gen)
+        res = eng.gnn_episode(jobs, explore=1.0, gen=gen, train=False)
+        dm, *_ = eng.actor_forward(jobs)
+        sp = eng.apsp(dm)
+        uds = torch.diagonal(dm.detach(), dim1=1, dim2=2)
+        dst, _ = eng.offload_decide(jobs, sp, uds, explore=1.0, gen=gen)
+        d = dst.cpu().numpy()
+        s = jobs.sources.cpu().numpy()
+        m = jobs.mask.cpu().numpy()
+        for b in range(eng.B):
+            for j in range(eng.Jmax):
+                if not m[b, j]:
+                    continue
+                total += 1
+                key = "local" if d[b, j] == s[b, j] else int(d[b, j])
+                counts[key] = counts.get(key, 0) + 1
+    # 3 servers + local = 4 choices, uniform → 25% each.  "local" also
+    # absorbs the rare greedy-pick-source case; generous tolerance.
+    assert abs(counts.get("local", 0) / total - 0.25) < 0.08
+    for s_ in (2, 3, 4):
+        assert abs(counts.get(s_, 0) / total - 0.25) < 0.08
+
+
+@needs_gpu
+def test_in_kernel_prob_mode_matches_torch_distribution():
+    """prob (softmax) sampling inside the kernel reproduces the torch
+    multinomial distribution over choices (statistical)."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    cases = _cases(B=8)
+    model = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    with torch.no_grad():
+        for p in model.parameters():
+            p.mul_(0.01)
+        model.layers[-1].bias.fill_(0.5)
+    eng_g = EpisodeEngine(cases, model, device="cuda", dtype=torch.float32)
+    model_c = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    with torch.no_grad():
+        for pc, pg in zip(model_c.parameters(), model.parameters()):
+            pc.copy_(pg)
+    eng_c = EpisodeEngine(cases, model_c, device="cpu", dtype=torch.float32)
+    eng_g.set_rng_seed(31337)
+    gen_g = torch.Generator(device="cuda")
+    gen_g.manual_seed(5)
+    gen_c = torch.Generator()
+    gen_c.manual_seed(5)
+
+    def dist(eng, gen, dev):
+        counts = {}
+        total = 0
+        for r in range(30):
+            g2 = torch.Generator(device=dev)
+            g2.manual_seed(1000 + r)
+            jobs = eng.sample_jobs(0.15, g2)
+            dm, *_ = eng.actor_forward(jobs)
+            sp = eng.apsp(dm)
+            uds = torch.diagonal(dm.detach(), dim1=1, dim2=2)
+            dst, _ = eng.offload_decide(jobs, sp, uds, gen=gen, prob=True)
+            d = dst.cpu().numpy()
+            s = jobs.sources.cpu().numpy()
+            m = jobs.mask.cpu().numpy()
+            for b in range(eng.B):
+                for j in range(eng.Jmax):
+                    if not m[b, j]:
+                        continue
+                    total += 1
+                    key = ("local" if d[b, j] == s[b, j]
+                           else int(d[b, j]))
+                    counts[key] = counts.get(key, 0) + 1
+        return {k: v / total for k, v in counts.items()}
+
+    pg = dist(eng_g, gen_g, "cuda")
+    pc = dist(eng_c, gen_c, "cpu")
+    keys = set(pg) | set(pc)
+    for k in keys:
+        assert abs(pg.get(k, 0.0) - pc.get(k, 0.0)) < 0.06, (k, pg, pc)
+
+
+@needs_gpu
+def test_delay_clamp_gpu_matches_cpu():
+    """delay_clamp threads identically through the HIP actor-head/critic
+    kernels and the CPU torch path (values AND gradients)."""
+    from multihop_offload_amd.engine import EpisodeEngine
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.graphs import JobInstance
+    from tests.test_engine import _jobbatch_from
+    cases = _cases()
+    mc = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    mg = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    with torch.no_grad():
+        for pc, pg in zip(mc.parameters(), mg.parameters()):
+            pc.mul_(0.01)
+            pg.copy_(pc)
+        mc.layers[-1].bias.fill_(0.5)
+        mg.layers[-1].bias.fill_(0.5)
+    eng_c = EpisodeEngine(cases, mc, device="cpu", dtype=torch.float32,
+                          delay_clamp=3.0)
+    eng_g = EpisodeEngine(cases, mg, device="cuda", dtype=torch.float32,
+                          delay_clamp=3.0)
+    insts = [JobInstance.sample(c.mobile_nodes, 0.6,
+                                np.random.RandomState(70 + i))
+             for i, c in enumerate(cases)]
+    res_c = eng_c.gnn_episode(_jobbatch_from(eng_c, insts), train=True)
+    res_g = eng_g.gnn_episode(_jobbatch_from(eng_g, insts), train=True)
+    assert np.allclose(res_c.tau.numpy(), res_g.tau.cpu().numpy(), rtol=1e-3)
+    for pc, pg in zip(mc.parameters(), mg.parameters()):
+        a, b = pc.grad.numpy(), pg.grad.cpu().numpy()
+        denom = max(np.abs(a).max(), 1e-6)
+        assert np.abs(a - b).max() / denom < 5e-3
+
+
+@needs_gpu
+def test_unit_mtx_deterministic_heterogeneous_jobs():
+    """Heterogeneous (ul+dl) jobs over shared congested links: the kernel's
+    packed atomicMax write order must reproduce the oracle's last-job-wins
+    unit matrix exactly (and be deterministic across repeats)."""
+    from multihop_offload_amd.engine import EpisodeEngine, JobBatch
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    cases = _cases(B=2)
+    mc = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    eng_c = EpisodeEngine(cases, mc, device="cpu", dtype=torch.float32)
+    eng_g = EpisodeEngine(cases, mc, device="cuda", dtype=torch.float32)
+
+    def het_jobs(eng, dev):
+        B, J = eng.B, eng.Jmax
+        # all jobs from the same mobile, heterogeneous ul → different
+        # fallback units on the SAME links under forced congestion
+        src = torch.full((B, J), int(cases[0].mobile_nodes[0]),
+                         dtype=torch.int64, device=dev)
+        mask = torch.ones(B, J, dtype=torch.bool, device=dev)
+        rates = torch.full((B, J), 5.0, device=dev)
+        ul = (50.0 + 10.0 * torch.arange(
+            J, device=dev, dtype=torch.float32)).expand(B, J).contiguous()
+        dl = torch.ones(B, J, device=dev)
+        return JobBatch(sources=src, mask=mask, rates=rates, ul=ul, dl=dl)
+
+    jb_c = het_jobs(eng_c, "cpu")
+    jb_g = het_jobs(eng_g, "cuda")
+    dst_c = torch.full_like(jb_c.sources, int(cases[0].servers[0]))
+    dst_g = dst_c.cuda()
+    sp_c, _, _ = eng_c.actor_forward(jb_c)
+    sp_c = eng_c.apsp(sp_c)
+    sp_g = sp_c.cuda()
+
+    _, _, de_c, um_c, wr_c = eng_c._episode_eval(jb_c, dst_c, sp_c)
+    um_gs = []
+    for _ in range(3):
+        _, _, de_g, um_g, wr_g = eng_g._episode_eval(jb_g, dst_g, sp_g)
+        um_gs.append(um_g.cpu())
+    assert torch.equal(um_gs[0], um_gs[1]) and torch.equal(um_gs[1],
+                                                           um_gs[2])
+    assert torch.equal(wr_c.cpu(), wr_g.cpu())
+    # last-job-wins parity with the oracle path
+    assert np.allclose(um_c.numpy(), um_gs[0].numpy(), rtol=1e-5, atol=1e-6)
+    assert np.allclose(de_c.numpy(), de_g.cpu().numpy(), rtol=1e-4)
+
+
+@needs_gpu
+def test_train_batched_capture_smoke(tmp_path):
+    """hipGraph-captured trainer: 30 steps with explore decay on a device
+    scalar; must finish, write history, and move parameters."""
+    from multihop_offload_amd.harness import train_batched
+    hist = train_batched.main([
+        "--steps", "30", "--batch", "16", "--nodes", "20",
+        "--distinct", "4", "--T", "500", "--workers", "0",
+        "--log_every", "10", "--save_every", "1000",
+        "--guard_every", "0", "--capture",
+        "--model_root", str(tmp_path), "--training_set", "CAPT"])
+    assert any("tau" in h for h in hist)

@@ -172,3 +172,35 @@ def test_fixed_point_properties_random():
         lam2[j] += 5.0
         mu2 = fixed_point_mu(lam2, rates, cf_degs, conf, 10)
         assert torch.all(mu2 <= mu + 1e-9), (mu2 - mu).max()
+
+
+def test_delay_fallback_cap_values_and_grad():
+    """delay_clamp: the 1/(mu-lam) branch is clamped at `cap` with
+    torch.clamp gradient semantics (zero where clamped); the congested
+    fallback branch is never clamped."""
+    from multihop_offload_amd.queueing import delay_with_fallback
+    lam = torch.tensor([1.0, 9.99, 12.0], dtype=torch.float64,
+                       requires_grad=True)
+    mu = torch.tensor([10.0, 10.0, 10.0], dtype=torch.float64)
+    out = delay_with_fallback(lam, mu, 1000.0, 101.0, cap=5.0)
+    # normal branch: 1/9 ≈ 0.111; near-pole branch: 1/0.01 = 100 → capped
+    assert torch.allclose(out[0], torch.tensor(1 / 9.0, dtype=torch.float64))
+    assert float(out[1]) == 5.0
+    # congested: fallback T*lam/(101*mu), NOT capped
+    assert torch.allclose(out[2],
+                          torch.tensor(1000.0 * 12.0 / (101.0 * 10.0),
+                                       dtype=torch.float64))
+    out.sum().backward()
+    g = lam.grad
+    assert abs(float(g[0]) - 1.0 / 81.0) < 1e-12   # d(1/(mu-lam))/dlam
+    assert float(g[1]) == 0.0                       # clamped → zero grad
+    assert abs(float(g[2]) - 1000.0 / (101.0 * 10.0)) < 1e-9
+
+
+def test_delay_fallback_cap_zero_is_reference():
+    from multihop_offload_amd.queueing import delay_with_fallback
+    lam = torch.rand(50, dtype=torch.float64) * 20
+    mu = torch.rand(50, dtype=torch.float64) * 20 + 0.1
+    a = delay_with_fallback(lam, mu, 1000.0, 101.0)
+    b = delay_with_fallback(lam, mu, 1000.0, 101.0, cap=0.0)
+    assert torch.equal(a, b)
