@@ -146,6 +146,156 @@ def read_bundle(prefix: str) -> Dict[str, np.ndarray]:
     return tensors
 
 
+# --------------------------------------------------------------------------
+# Bundle WRITER — emits a TF-format checkpoint (.index SSTable +
+# .data-00000-of-00001 shard) from a ChebConvStack, with the reference's
+# key layout and TF-masked crc32c per tensor, so the round-trip back into
+# the reference's TF stack (gnn_offloading_agent.py:125-132 /
+# tf.train.load_checkpoint) is possible without TF installed here.
+# --------------------------------------------------------------------------
+
+_CRC32C_TABLE = None
+
+
+def _crc32c(data: bytes) -> int:
+    """CRC-32C (Castagnoli), as TF's tensor-bundle uses."""
+    global _CRC32C_TABLE
+    if _CRC32C_TABLE is None:
+        poly = 0x82F63B78
+        tbl = []
+        for i in range(256):
+            c = i
+            for _ in range(8):
+                c = (c >> 1) ^ poly if c & 1 else c >> 1
+            tbl.append(c)
+        _CRC32C_TABLE = tbl
+    crc = 0xFFFFFFFF
+    for b in data:
+        crc = _CRC32C_TABLE[(crc ^ b) & 0xFF] ^ (crc >> 8)
+    return crc ^ 0xFFFFFFFF
+
+
+def _masked_crc32c(data: bytes) -> int:
+    c = _crc32c(data)
+    return ((c >> 15) | (c << 17)) + 0xA282EAD8 & 0xFFFFFFFF
+
+
+def _enc_varint(v: int) -> bytes:
+    out = b""
+    while True:
+        b = v & 0x7F
+        v >>= 7
+        if v:
+            out += bytes([b | 0x80])
+        else:
+            return out + bytes([b])
+
+
+def _enc_block(entries) -> bytes:
+    """One SSTable block, no prefix compression (shared=0 every entry,
+    single restart point)."""
+    body = b""
+    for key, value in entries:
+        kb = key.encode()
+        body += _enc_varint(0) + _enc_varint(len(kb)) \
+            + _enc_varint(len(value)) + kb + value
+    body += struct.pack("<I", 0)        # restart offset 0
+    body += struct.pack("<I", 1)        # n_restarts
+    return body
+
+
+def _block_trailer(block: bytes) -> bytes:
+    # compression byte 0 + masked crc32c over (block || type byte)
+    return b"\x00" + struct.pack("<I", _masked_crc32c(block + b"\x00"))
+
+
+def _entry_proto(dtype_code: int, shape, offset: int, size: int,
+                 crc: int) -> bytes:
+    shp = b""
+    for d in shape:
+        dim = _enc_varint(1 << 3 | 0) + _enc_varint(d)   # Dim.size
+        shp += _enc_varint(2 << 3 | 2) + _enc_varint(len(dim)) + dim
+    out = _enc_varint(1 << 3 | 0) + _enc_varint(dtype_code)
+    out += _enc_varint(2 << 3 | 2) + _enc_varint(len(shp)) + shp
+    if offset:
+        out += _enc_varint(4 << 3 | 0) + _enc_varint(offset)
+    out += _enc_varint(5 << 3 | 0) + _enc_varint(size)
+    out += _enc_varint(6 << 3 | 5) + struct.pack("<I", crc)
+    return out
+
+
+def write_bundle(prefix: str, tensors: Dict[str, np.ndarray]):
+    """Write `{key: array}` as a single-shard TF bundle at `prefix`
+    (``prefix.index`` + ``prefix.data-00000-of-00001``)."""
+    np_to_code = {np.dtype(np.float32): 1, np.dtype(np.float64): 2,
+                  np.dtype(np.int32): 3, np.dtype(np.int64): 9}
+    keys = sorted(tensors)
+    shard = b""
+    entries = []
+    # header entry (key ""): BundleHeaderProto {num_shards=1, version{producer=1}}
+    ver = _enc_varint(1 << 3 | 0) + _enc_varint(1)
+    header = _enc_varint(1 << 3 | 0) + _enc_varint(1) \
+        + _enc_varint(3 << 3 | 2) + _enc_varint(len(ver)) + ver
+    entries.append(("", header))
+    for k in keys:
+        a = np.ascontiguousarray(tensors[k])
+        code = np_to_code[a.dtype]
+        data = a.tobytes()
+        entries.append((k, _entry_proto(code, a.shape, len(shard),
+                                        len(data), _masked_crc32c(data))))
+        shard += data
+
+    data_block = _enc_block(entries)
+    index = data_block + _block_trailer(data_block)
+    data_handle = _enc_varint(0) + _enc_varint(len(data_block))
+
+    meta_off = len(index)
+    meta_block = _enc_block([])
+    index += meta_block + _block_trailer(meta_block)
+    meta_handle = _enc_varint(meta_off) + _enc_varint(len(meta_block))
+
+    ix_off = len(index)
+    last_key = keys[-1] if keys else ""
+    ix_block = _enc_block([(last_key, data_handle)])
+    index += ix_block + _block_trailer(ix_block)
+    ix_handle = _enc_varint(ix_off) + _enc_varint(len(ix_block))
+
+    footer = meta_handle + ix_handle
+    footer += b"\x00" * (40 - len(footer))
+    footer += struct.pack("<Q", _MAGIC)
+    index += footer
+
+    with open(prefix + ".index", "wb") as f:
+        f.write(index)
+    with open(prefix + ".data-00000-of-00001", "wb") as f:
+        f.write(shard)
+
+
+def save_reference_weights(model, prefix: str, dtype=np.float64):
+    """Write a ``ChebConvStack``'s weights as a TF bundle with the
+    reference's key layout (``layer_with_weights-<i>/{kernel,bias}/
+    .ATTRIBUTES/VARIABLE_VALUE``, float64 like the reference's Keras
+    model), plus a ``checkpoint`` manifest so
+    ``tf.train.latest_checkpoint`` finds it.  Name-based TF restore
+    (``tf.train.load_checkpoint``) reads it directly; Keras
+    object-graph-based ``load_weights`` additionally wants the
+    ``_CHECKPOINTABLE_OBJECT_GRAPH`` entry, which name-based loaders
+    ignore — use ``tf.train.load_checkpoint`` on the TF side."""
+    tensors = {}
+    for i, layer in enumerate(model.layers):
+        base = f"layer_with_weights-{i}"
+        tensors[f"{base}/kernel/.ATTRIBUTES/VARIABLE_VALUE"] = \
+            layer.weight.detach().cpu().numpy().astype(dtype)
+        tensors[f"{base}/bias/.ATTRIBUTES/VARIABLE_VALUE"] = \
+            layer.bias.detach().cpu().numpy().astype(dtype)
+    write_bundle(prefix, tensors)
+    import os
+    d, name = os.path.split(prefix)
+    with open(os.path.join(d or ".", "checkpoint"), "w") as f:
+        f.write(f'model_checkpoint_path: "{name}"\n'
+                f'all_model_checkpoint_paths: "{name}"\n')
+
+
 def load_reference_weights(model, prefix: str):
     """Load a reference Keras checkpoint into a ``ChebConvStack``.  Keys:
     ``layer_with_weights-<i>/{kernel,bias}/.ATTRIBUTES/VARIABLE_VALUE``.

@@ -675,8 +675,9 @@ def test_unit_mtx_deterministic_heterogeneous_jobs():
     from multihop_offload_amd.models.chebconv import ChebConvStack
     cases = _cases(B=2)
     mc = ChebConvStack(K=2, dtype=torch.float32, seed=3)
+    mg = ChebConvStack(K=2, dtype=torch.float32, seed=3)
     eng_c = EpisodeEngine(cases, mc, device="cpu", dtype=torch.float32)
-    eng_g = EpisodeEngine(cases, mc, device="cuda", dtype=torch.float32)
+    eng_g = EpisodeEngine(cases, mg, device="cuda", dtype=torch.float32)
 
     def het_jobs(eng, dev):
         B, J = eng.B, eng.Jmax
