@@ -81,7 +81,7 @@ class EpisodeEngine:
     def __init__(self, cases: Sequence[CaseGraph], model: ChebConvStack,
                  device: str = "cpu", dtype: torch.dtype = torch.float32,
                  fp_iters: int = 10, walk_cap: Optional[int] = None,
-                 delay_clamp: float = 0.0):
+                 delay_clamp: float = 0.0, lam_margin: float = 0.0):
         assert len(cases) > 0
         N = cases[0].num_nodes
         T = cases[0].T
@@ -101,6 +101,11 @@ class EpisodeEngine:
         # pole mitigation for training: clamp the 1/(mu-lam) delay branch
         # at this value (0 = off, reference semantics) — see queueing.py
         self.delay_clamp = float(delay_clamp)
+        # inference-time conservatism: the decision stage sees predicted
+        # traffic inflated by (1+lam_margin) — a calibratable knob that
+        # biases marginal offloads toward local compute (congestion-tail
+        # control without retraining); 0 = reference behavior
+        self.lam_margin = float(lam_margin)
         self.model = model.to(self.device)
         self.walk_cap = walk_cap or N
 
@@ -436,6 +441,8 @@ class EpisodeEngine:
                 for layer in self.model.layers:
                     params += [layer.weight, layer.bias]
                 lam = ChebStackFn.apply(x.contiguous(), self, *params)
+                if self.lam_margin:
+                    lam = lam * (1.0 + self.lam_margin)
             else:
                 # no silent eager fallback on GPU: unsupported model
                 # shapes refuse loudly unless explicitly overridden
@@ -452,6 +459,8 @@ class EpisodeEngine:
             return dm, None, None
 
         lam = self.model(x.reshape(B * Ee, 4), self.support).reshape(B, Ee)
+        if self.lam_margin:
+            lam = lam * (1.0 + self.lam_margin)
 
         lam_link = lam[:, :E].reshape(-1)
         mu = fixed_point_mu(lam_link, self.link_rates.reshape(-1),

@@ -24,7 +24,7 @@ from .train_batched import build_training_cases
 
 
 def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
-             device, dtype, workers: int = 8):
+             device, dtype, workers: int = 8, lam_margin: float = 0.0):
     """Returns per-method aggregate {tau, congest_jobs, num_jobs} summed /
     averaged over all (size, case, instance)."""
     agg = {m: {"tau_sum": 0.0, "tau_n": 0, "congest": 0, "jobs": 0,
@@ -34,7 +34,8 @@ def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
     for n in sizes:
         cases = build_training_cases(n, cases_per_size, cases_per_size, T,
                                      seed + n, workers=workers)
-        engine = EpisodeEngine(cases, model, device=device, dtype=dtype)
+        engine = EpisodeEngine(cases, model, device=device, dtype=dtype,
+                               lam_margin=lam_margin)
         gen = torch.Generator(device=device)
         gen.manual_seed(seed + n)
         ps = {m: {"tau_sum": 0.0, "tau_n": 0, "congest": 0, "jobs": 0,
@@ -87,6 +88,10 @@ def main(argv=None):
     ap.add_argument("--seed", type=int, default=500)
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--workers", type=int, default=8)
+    ap.add_argument("--lam_margin", type=float, default=0.0,
+                    help="inference-time conservatism: inflate predicted "
+                         "traffic by (1+margin) at decision time "
+                         "(congestion-tail control; 0 = reference)")
     ap.add_argument("--out", type=str, default="out/eval_summary.json")
     args = ap.parse_args(argv)
 
@@ -103,7 +108,8 @@ def main(argv=None):
     sizes = [int(s) for s in args.sizes.split(",")]
     summary, per_size = evaluate(model, sizes, args.cases_per_size,
                                  args.instances, args.T, args.load,
-                                 args.seed, device, dtype, args.workers)
+                                 args.seed, device, dtype, args.workers,
+                                 lam_margin=args.lam_margin)
     os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
     with open(args.out, "w") as f:
         json.dump({"summary": summary, "per_size": per_size,

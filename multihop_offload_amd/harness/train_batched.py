@@ -96,11 +96,12 @@ def _dist_mean(x: float, world: int, device) -> float:
     return float(x)
 
 
-def evaluate_policy(engines, loads, seed: int, rounds: int = 4) -> float:
-    """Held-out evaluation: mean per-job tau of the greedy policy
-    (explore=0, no gradients) over fresh job draws from a FIXED seed, so
-    successive calls during training are comparable."""
+def evaluate_policy(engines, loads, seed: int, rounds: int = 4):
+    """Held-out evaluation: mean per-job tau AND congestion ratio of the
+    greedy policy (explore=0, no gradients) over fresh job draws from a
+    FIXED seed, so successive calls during training are comparable."""
     taus = []
+    congest = jobs_n = 0
     with torch.no_grad():
         for engine in engines:
             gen = torch.Generator(device=engine.device)
@@ -109,7 +110,10 @@ def evaluate_policy(engines, loads, seed: int, rounds: int = 4) -> float:
                 jobs = engine.sample_jobs(loads[r % len(loads)], gen)
                 res = engine.gnn_episode(jobs, train=False)
                 taus.append(res.tau.flatten())
-    return float(torch.nanmean(torch.cat(taus)))
+                congest += int(res.congest.sum())
+                jobs_n += int(res.num_jobs.sum())
+    return (float(torch.nanmean(torch.cat(taus))),
+            congest / max(jobs_n, 1))
 
 
 def main(argv=None):
@@ -166,6 +170,11 @@ def main(argv=None):
                          "the default held-out ones (held-out catches "
                          "topology overfitting — see docs/TRAINING.md)")
     ap.add_argument("--eval_seed", type=int, default=12345)
+    ap.add_argument("--eval_congest_weight", type=float, default=3000.0,
+                    help="held-out selection metric = eval tau + this "
+                         "weight x eval congestion ratio (the north-star "
+                         "quality clause is the congestion TAIL, which "
+                         "tau alone does not rank); 0 reverts to pure tau")
     ap.add_argument("--torch_profile", type=int, default=0,
                     help="profile N steps with torch.profiler after warmup "
                          "and write a chrome trace next to the model dir")
@@ -374,16 +383,19 @@ def main(argv=None):
                 group["lr"] *= 0.1
             hip_graphs.clear()      # lr is baked into captured graphs
         if args.eval_every and step % args.eval_every == 0:
-            eval_tau = _dist_mean(
-                evaluate_policy(eval_engines, loads, args.eval_seed,
-                                args.eval_rounds),
-                world, engine.device)
-            if eval_tau < best_eval_tau:
-                best_eval_tau = eval_tau
+            et, ec = evaluate_policy(eval_engines, loads, args.eval_seed,
+                                     args.eval_rounds)
+            eval_tau = _dist_mean(et, world, engine.device)
+            eval_congest = _dist_mean(ec, world, engine.device)
+            eval_metric = eval_tau + args.eval_congest_weight * eval_congest
+            if eval_metric < best_eval_tau:
+                best_eval_tau = eval_metric
                 best_eval_params = [p.detach().clone()
                                     for p in engine.model.parameters()]
             rec = {"step": step, "eval_tau": eval_tau,
-                   "best_eval_tau": best_eval_tau}
+                   "eval_congest": eval_congest,
+                   "eval_metric": eval_metric,
+                   "best_eval_metric": best_eval_tau}
             history.append(rec)        # every rank: values are all-reduced
             if rank == 0:
                 print(json.dumps(rec), flush=True)
