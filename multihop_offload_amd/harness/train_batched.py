@@ -142,8 +142,10 @@ def main(argv=None):
     ap.add_argument("--save_every", type=int, default=500)
     ap.add_argument("--log_every", type=int, default=50)
     ap.add_argument("--workers", type=int, default=8)
-    ap.add_argument("--lr_decay_at", type=int, default=0,
-                    help="step after which lr is multiplied by 0.1")
+    ap.add_argument("--lr_decay_at", type=str, default="0",
+                    help="step(s) after which lr is multiplied by 0.1 "
+                         "(comma-separated for multi-decay schedules, "
+                         "e.g. 30000,70000; 0 disables)")
     ap.add_argument("--guard_every", type=int, default=1000,
                     help="divergence guard period (0 disables): if the "
                          "running tau explodes past 5x the best seen, roll "
@@ -271,6 +273,8 @@ def main(argv=None):
                                  workers=args.workers),
             model, device=device, dtype=dtype) for n in sizes]
 
+    lr_decay_steps = {int(x) for x in str(args.lr_decay_at).split(",")
+                      if int(x or 0) > 0}
     actor_dir = model_dir(args.model_root, args.training_set)
     explore = args.explore
     t0 = time.time()
@@ -378,7 +382,7 @@ def main(argv=None):
             opt.step()
             engine.model.apply_constraints()
         explore = max(explore * args.explore_decay, 0.001)
-        if args.lr_decay_at and step == args.lr_decay_at:
+        if step in lr_decay_steps:
             for group in opt.param_groups:
                 group["lr"] *= 0.1
             hip_graphs.clear()      # lr is baked into captured graphs
