@@ -220,11 +220,30 @@ DEV_INLINE void spmv(const float* __restrict__ src, float* __restrict__ dst,
         const int c0 = (task % (F / 8)) * 8;
         float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
         if (r < Ee) {
-            for (int a = indptr[r]; a < indptr[r + 1]; ++a) {
+            // 4-way unrolled neighbor loop: 4 independent row pointers keep
+            // 32 LDS reads in flight per iteration instead of 8 (the serial
+            // version is latency-bound on the ds_read chain)
+            int a = indptr[r];
+            const int end = indptr[r + 1];
+            float acc1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+            for (; a + 3 < end; a += 4) {
+                const float* s0 = src + cols[a] * STRIDE + c0;
+                const float* s1 = src + cols[a + 1] * STRIDE + c0;
+                const float* s2 = src + cols[a + 2] * STRIDE + c0;
+                const float* s3 = src + cols[a + 3] * STRIDE + c0;
+#pragma unroll
+                for (int c = 0; c < 8; ++c) {
+                    acc[c] += s0[c] + s1[c];
+                    acc1[c] += s2[c] + s3[c];
+                }
+            }
+            for (; a < end; ++a) {
                 const float* s = src + cols[a] * STRIDE + c0;
 #pragma unroll
                 for (int c = 0; c < 8; ++c) acc[c] += s[c];
             }
+#pragma unroll
+            for (int c = 0; c < 8; ++c) acc[c] += acc1[c];
         }
         float* d = dst + r * STRIDE + c0;
         if (mode == 0) {

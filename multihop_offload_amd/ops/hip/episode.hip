@@ -357,8 +357,11 @@ std::vector<torch::Tensor> walk_eval_hip(
     auto overflow = torch::zeros({B}, opts_i.dtype(torch::kInt32));
     const size_t lds = sizeof(float) * (3 * (size_t)E + N);
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for LDS walk_eval");
+    // large graphs ship few workgroups (one per graph): widen them so the
+    // E-length loops and the N*N unpack keep more lanes busy per CU
+    const int threads = (E >= 1500 || N >= 500) ? 1024 : 256;
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(walk_eval_kernel, dim3(B), dim3(256), lds,
+    hipLaunchKernelGGL(walk_eval_kernel, dim3(B), dim3(threads), lds,
                        stream.stream(),
                        sp.data_ptr<float>(), src.data_ptr<long>(),
                        dst.data_ptr<long>(), mask.data_ptr<bool>(),

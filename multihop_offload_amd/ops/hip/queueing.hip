@@ -452,8 +452,9 @@ std::vector<torch::Tensor> critic_hip(
         g_dunit = g_hist;
         g_dlam = g_hist;
     }
+    const int threads = E >= 1500 ? 1024 : 256;
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(critic_kernel, dim3(B), dim3(256), lds,
+    hipLaunchKernelGGL(critic_kernel, dim3(B), dim3(threads), lds,
                        stream.stream(),
                        route_links.data_ptr<int>(), nhop.data_ptr<int>(),
                        vedge_dst.data_ptr<long>(), mask.data_ptr<bool>(),
@@ -488,8 +489,9 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
         lds = sizeof(float) * 2 * (size_t)E;
     }
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for LDS actor head");
+    const int threads = E >= 1500 ? 1024 : 256;
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(actor_head_fwd_kernel, dim3(B), dim3(256), lds,
+    hipLaunchKernelGGL(actor_head_fwd_kernel, dim3(B), dim3(threads), lds,
                        stream.stream(),
                        lam_ext.data_ptr<float>(),
                        conf_indptr.data_ptr<int>(),
@@ -520,8 +522,9 @@ torch::Tensor actor_head_bwd_hip(
         lds = sizeof(float) * 5 * (size_t)E;
     }
     TORCH_CHECK(lds <= 160 * 1024, "graph too large for LDS actor head bwd");
+    const int threads = E >= 1500 ? 1024 : 256;
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(actor_head_bwd_kernel, dim3(B), dim3(256), lds,
+    hipLaunchKernelGGL(actor_head_bwd_kernel, dim3(B), dim3(threads), lds,
                        stream.stream(),
                        grad_dist.data_ptr<float>(),
                        lam_ext.data_ptr<float>(), mu_hist.data_ptr<float>(),
