@@ -172,6 +172,13 @@ def main(argv=None):
                          "the default held-out ones (held-out catches "
                          "topology overfitting — see docs/TRAINING.md)")
     ap.add_argument("--eval_seed", type=int, default=12345)
+    ap.add_argument("--eval_guard", type=float, default=3.0,
+                    help="held-out divergence guard: if the eval metric "
+                         "exceeds this multiple of the best seen, roll "
+                         "back to the best-eval parameters and cut lr 3x "
+                         "(0 disables).  Catches the bistable collapse "
+                         "that stays invisible on TRAINING topologies "
+                         "(train tau ~17 while held-out tau >100)")
     ap.add_argument("--eval_congest_weight", type=float, default=3000.0,
                     help="held-out selection metric = eval tau + this "
                          "weight x eval congestion ratio (the north-star "
@@ -288,6 +295,8 @@ def main(argv=None):
     rollbacks = 0
     best_eval_tau = float("inf")
     best_eval_params = None
+    best_eval_opt = None
+    eval_rollbacks = 0
     loads = ([float(x) for x in args.arrival_scales.split(",")]
              if args.arrival_scales else [args.arrival_scale])
     profiler = None
@@ -392,14 +401,36 @@ def main(argv=None):
             eval_tau = _dist_mean(et, world, engine.device)
             eval_congest = _dist_mean(ec, world, engine.device)
             eval_metric = eval_tau + args.eval_congest_weight * eval_congest
+            rolled = 0
             if eval_metric < best_eval_tau:
                 best_eval_tau = eval_metric
                 best_eval_params = [p.detach().clone()
                                     for p in engine.model.parameters()]
+                best_eval_opt = _copy.deepcopy(opt.state_dict())
+            elif (args.eval_guard and best_eval_params is not None
+                  and eval_metric > args.eval_guard
+                  * max(best_eval_tau, 20.0)):
+                # held-out collapse: restore the best-eval state and cool
+                # the lr — turns the bistable divergence into a recoverable
+                # excursion instead of a lost run
+                with torch.no_grad():
+                    for p, bp in zip(engine.model.parameters(),
+                                     best_eval_params):
+                        p.copy_(bp)
+                if best_eval_opt is not None:
+                    opt.load_state_dict(best_eval_opt)
+                for group in opt.param_groups:
+                    group["lr"] /= 3.0
+                hip_graphs.clear()
+                eval_rollbacks += 1
+                rolled = 1
             rec = {"step": step, "eval_tau": eval_tau,
                    "eval_congest": eval_congest,
                    "eval_metric": eval_metric,
-                   "best_eval_metric": best_eval_tau}
+                   "best_eval_metric": best_eval_tau,
+                   "eval_rollbacks": eval_rollbacks,
+                   "lr": opt.param_groups[0]["lr"],
+                   "rolled": rolled}
             history.append(rec)        # every rank: values are all-reduced
             if rank == 0:
                 print(json.dumps(rec), flush=True)
