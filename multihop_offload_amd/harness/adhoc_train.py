@@ -37,7 +37,8 @@ def main(argv=None):
     except Exception:
         print(f"unable to load {actor_model}")
 
-    names = common.list_cases(args.datapath, args.limit_cases)
+    names = common.list_cases(args.datapath, args.limit_cases,
+                              getattr(args, 'shard', ''))
     output_csv = os.path.join(
         args.out, "aco_training_data_{}_load_{:.2f}_T_{}.csv".format(
             os.path.basename(os.path.normpath(args.datapath)),

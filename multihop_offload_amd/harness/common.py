@@ -52,6 +52,9 @@ def build_parser(default_datapath="../data_100") -> argparse.ArgumentParser:
         "reference when omitted)")
     add("--limit_cases", type=int, default=0,
         help="process at most this many .mat cases (0 = all)")
+    add("--shard", type=str, default="",
+        help="'I:K' — process cases I::K only (parallel sharding of a "
+             "large run; merge the per-shard CSVs afterwards)")
     return p
 
 
@@ -106,11 +109,14 @@ def run_method(method: str, agent: ACOAgent, env: AdhocCloudEnv,
     return delay_empirical(dl_links, dl_nodes)
 
 
-def list_cases(datapath: str, limit: int = 0):
+def list_cases(datapath: str, limit: int = 0, shard: str = ""):
     names = sorted(os.listdir(datapath))
     names = [n for n in names if n.endswith(".mat")]
     if limit:
         names = names[:limit]
+    if shard:
+        i, k = (int(x) for x in shard.split(":"))
+        names = names[i::k]
     return names
 
 
