@@ -292,6 +292,12 @@ class EpisodeEngine:
             cip[b, Eb + 1:] = c.conf_indptr[-1]
         self.k_conf_indptr = t32(cip)
         self.k_E_arr = t32([c.num_links for c in cases])           # (B,)
+        # per-graph effective node counts (pad_to batches carry inert pad
+        # nodes; kernels skip them — O(n³) APSP instead of O(Nmax³))
+        self.k_N_arr = t32([getattr(c, "real_n", c.num_nodes)
+                            for c in cases])
+        self._padded = any(getattr(c, "real_n", c.num_nodes) != N
+                           for c in cases)
         cc, base = [], [0]
         for c in cases:
             cc.append(np.asarray(c.conf_indices, dtype=np.int32))
@@ -503,7 +509,8 @@ class EpisodeEngine:
             w = torch.where(self.adj, dm.detach(),
                             torch.full_like(dm, float("inf")))
             w = torch.where(self._eye, torch.zeros_like(w), w)
-            return ops.floyd_warshall(w)
+            return ops.floyd_warshall(
+                w, self.k_N_arr if self._padded else None)
 
     # ----------------------------------------------- decision + routing walk
     def offload_decide(self, jobs: JobBatch, sp: torch.Tensor,
@@ -648,7 +655,8 @@ class EpisodeEngine:
                 self.k_adj_link, self.k_conf_indptr, self.k_conf_base,
                 self.k_conf_cols, self.link_rates.contiguous(),
                 self.proc_bws.contiguous(), self.k_edges,
-                self.T_arr.contiguous(), H, self.fp_iters)
+                self.T_arr.contiguous(), self.k_E_arr, self.k_N_arr, H,
+                self.fp_iters)
             self._last_overflow = overflow
             self.overflow_total += overflow.sum().to(torch.int64)
             return rl, nhop, delay_emp, unit_mtx, written
@@ -775,7 +783,8 @@ class EpisodeEngine:
                 jobs.rates.contiguous(), jobs.ul.contiguous(),
                 jobs.dl.contiguous(), self.k_conf_indptr, self.k_conf_base,
                 self.k_conf_cols, self.link_rates.contiguous(),
-                self.bw_comp.contiguous(), self.T_arr.contiguous(), Ee,
+                self.bw_comp.contiguous(), self.k_E_arr,
+                self.T_arr.contiguous(), Ee,
                 self.fp_iters, self.delay_clamp)
             return grad_edge, loss.sum()
         H = route_links.shape[2]

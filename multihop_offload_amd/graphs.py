@@ -294,6 +294,7 @@ class CaseGraph:
         g = CaseGraph(n_target, t_max=self.T, seed=self.seed, m=self.m,
                       gtype=self.gtype, adj=adj, pos=pos,
                       cf_radius=self.cf_radius)
+        g.real_n = N0          # kernels bound their loops at the real size
         for v in range(N0):
             if self.roles[v] == 2:
                 g.add_relay(v)

@@ -48,8 +48,11 @@ def require_hip():
 _require_hip = require_hip
 
 
-def floyd_warshall(w: torch.Tensor) -> torch.Tensor:
+def floyd_warshall(w: torch.Tensor, n_arr=None) -> torch.Tensor:
+    """Batched min-plus APSP.  ``n_arr`` (optional (B,) int32): per-graph
+    effective node count for inert-padded batches — the kernel relaxes
+    only the leading n×n submatrix (pad rows are +inf already)."""
     if w.is_cuda and os.environ.get("MHO_FORCE_TORCH") != "1":
         ext = _require_hip()
-        return ext.floyd_warshall(w.contiguous())
+        return ext.floyd_warshall(w.contiguous(), n_arr)
     return torch_ref.floyd_warshall(w)

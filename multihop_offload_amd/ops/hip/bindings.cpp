@@ -1,7 +1,8 @@
 // Python bindings for the gfx950 HIP kernels.
 #include <torch/extension.h>
 
-torch::Tensor floyd_warshall_hip(torch::Tensor w);
+torch::Tensor floyd_warshall_hip(torch::Tensor w,
+                                 c10::optional<torch::Tensor> n_arr);
 std::vector<torch::Tensor> decide_hip(
     torch::Tensor sp, torch::Tensor hop, torch::Tensor uds,
     torch::Tensor servers, torch::Tensor src, torch::Tensor mask,
@@ -14,14 +15,16 @@ std::vector<torch::Tensor> walk_eval_hip(
     torch::Tensor dl, torch::Tensor adj_indptr, torch::Tensor adj_idx,
     torch::Tensor adj_link, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
-    torch::Tensor bw, torch::Tensor edges, torch::Tensor T_arr, long H,
+    torch::Tensor bw, torch::Tensor edges, torch::Tensor T_arr,
+    torch::Tensor E_arr, torch::Tensor n_arr, long H,
     long fp_iters);
 std::vector<torch::Tensor> critic_hip(
     torch::Tensor route_links, torch::Tensor nhop, torch::Tensor vedge_dst,
     torch::Tensor mask, torch::Tensor rate, torch::Tensor ul,
     torch::Tensor dl, torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
-    torch::Tensor T_arr, long Ee, long iters, double cap);
+    torch::Tensor E_arr, torch::Tensor T_arr, long Ee, long iters,
+    double cap);
 std::vector<torch::Tensor> actor_head_fwd_hip(
     torch::Tensor lam_ext, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,

@@ -187,6 +187,8 @@ __global__ void walk_eval_kernel(
     unsigned long long* __restrict__ upack,  // (B,N,N) scratch
     int* __restrict__ overflow,          // (B) out
     const float* __restrict__ T_arr,     // (B)
+    const int* __restrict__ E_arr,       // (B) real link counts
+    const int* __restrict__ n_arr,       // (B) real node counts (padding)
     int N, int E, int J, int H, int fp_iters) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* lam = reinterpret_cast<float*>(smem_raw);          // E
@@ -206,10 +208,16 @@ __global__ void walk_eval_kernel(
     const int* ccols = conf_cols + conf_base[b];
     const float T = T_arr[b];
     unsigned long long* upk = upack + (size_t)b * N * N;
+    // per-graph effective sizes: padded link slots / inert pad nodes are
+    // isolated and never touched by routes, so every E/N-length loop below
+    // runs on the real extent only (ragged-E batches, pad_to mixed-N)
+    const int Eb = E_arr[b];
+    const int nb_ = n_arr[b];
 
-    for (int e = tid; e < E; e += nt) lam[e] = 0.0f;
-    for (int n = tid; n < N; n += nt) sload[n] = 0.0f;
-    for (int i = tid; i < N * N; i += nt) upk[i] = 0ull;
+    for (int e = tid; e < Eb; e += nt) lam[e] = 0.0f;
+    for (int n = tid; n < nb_; n += nt) sload[n] = 0.0f;
+    for (int i = tid; i < nb_ * nb_; i += nt)
+        upk[(i / nb_) * N + (i % nb_)] = 0ull;
     __syncthreads();
 
     // ---- stage 1: walk + load accumulation -------------------------------
@@ -241,18 +249,18 @@ __global__ void walk_eval_kernel(
     __syncthreads();
 
     // ---- stage 2: contention fixed point (offloading_v3.py:500-506) ------
-    for (int e = tid; e < E; e += nt) {
+    for (int e = tid; e < Eb; e += nt) {
         const float deg = (float)(cip[e + 1] - cip[e]);
         mu[e] = ratesb[e] / (deg + 1.0f);
     }
     __syncthreads();
     for (int it = 0; it < fp_iters; ++it) {
-        for (int e = tid; e < E; e += nt) {
+        for (int e = tid; e < Eb; e += nt) {
             const float r = lam[e] / mu[e];
             busy[e] = r < 0.f ? 0.f : (r > 1.f ? 1.f : r);
         }
         __syncthreads();
-        for (int e = tid; e < E; e += nt) {
+        for (int e = tid; e < Eb; e += nt) {
             float nbv = 0.0f;
             for (int a = cip[e]; a < cip[e + 1]; ++a) nbv += busy[ccols[a]];
             mu[e] = ratesb[e] / (1.0f + nbv);
@@ -297,11 +305,12 @@ __global__ void walk_eval_kernel(
     // ---- stage 3b: unpack last-job-wins units into unit_mtx/written ------
     float* um = unit_mtx + (size_t)b * N * N;
     bool* wm = written + (size_t)b * N * N;
-    for (int i = tid; i < N * N; i += nt) {
-        const unsigned long long pk = upk[i];
+    for (int i = tid; i < nb_ * nb_; i += nt) {
+        const int cell = (i / nb_) * N + (i % nb_);
+        const unsigned long long pk = upk[cell];
         if (pk) {
-            um[i] = __uint_as_float((unsigned int)(pk & 0xffffffffull));
-            wm[i] = true;
+            um[cell] = __uint_as_float((unsigned int)(pk & 0xffffffffull));
+            wm[cell] = true;
         }
     }
 }
@@ -339,7 +348,8 @@ std::vector<torch::Tensor> walk_eval_hip(
     torch::Tensor dl, torch::Tensor adj_indptr, torch::Tensor adj_idx,
     torch::Tensor adj_link, torch::Tensor conf_indptr,
     torch::Tensor conf_base, torch::Tensor conf_cols, torch::Tensor rates,
-    torch::Tensor bw, torch::Tensor edges, torch::Tensor T_arr, long H,
+    torch::Tensor bw, torch::Tensor edges, torch::Tensor T_arr,
+    torch::Tensor E_arr, torch::Tensor n_arr, long H,
     long fp_iters) {
     const int B = sp.size(0), N = sp.size(1);
     const int J = src.size(1), E = rates.size(1);
@@ -379,7 +389,8 @@ std::vector<torch::Tensor> walk_eval_hip(
                        reinterpret_cast<unsigned long long*>(
                            upack.data_ptr<long>()),
                        overflow.data_ptr<int>(),
-                       T_arr.data_ptr<float>(), N, E, J, (int)H,
+                       T_arr.data_ptr<float>(), E_arr.data_ptr<int>(),
+                       n_arr.data_ptr<int>(), N, E, J, (int)H,
                        (int)fp_iters);
     return {route_links, nhop, delay_emp, unit_mtx, written, overflow};
 }

@@ -23,24 +23,26 @@ namespace {
 #define DEV_INLINE __device__ __forceinline__
 
 // forward fixed point storing mu_t for t=0..iters into `hist` ((iters+1)*E)
+// Eb: real (per-graph) link count the loops run over; Estride: row stride
+// of the mu history (the BATCH-max E — ragged-E / padded batches)
 DEV_INLINE void fixed_point_fwd(const float* lam, const float* rates,
                                 const int* cip, const int* ccols,
-                                float* hist, float* busy, int E, int iters,
-                                int tid, int nt) {
-    for (int e = tid; e < E; e += nt) {
+                                float* hist, float* busy, int Eb,
+                                int Estride, int iters, int tid, int nt) {
+    for (int e = tid; e < Eb; e += nt) {
         const float deg = (float)(cip[e + 1] - cip[e]);
         hist[e] = rates[e] / (deg + 1.0f);
     }
     __syncthreads();
     for (int t = 1; t <= iters; ++t) {
-        const float* mu_prev = hist + (size_t)(t - 1) * E;
-        float* mu_cur = hist + (size_t)t * E;
-        for (int e = tid; e < E; e += nt) {
+        const float* mu_prev = hist + (size_t)(t - 1) * Estride;
+        float* mu_cur = hist + (size_t)t * Estride;
+        for (int e = tid; e < Eb; e += nt) {
             const float r = lam[e] / mu_prev[e];
             busy[e] = r < 0.f ? 0.f : (r > 1.f ? 1.f : r);
         }
         __syncthreads();
-        for (int e = tid; e < E; e += nt) {
+        for (int e = tid; e < Eb; e += nt) {
             float nbv = 0.0f;
             for (int a = cip[e]; a < cip[e + 1]; ++a) nbv += busy[ccols[a]];
             mu_cur[e] = rates[e] / (1.0f + nbv);
@@ -54,24 +56,24 @@ DEV_INLINE void fixed_point_fwd(const float* lam, const float* rates,
 DEV_INLINE void fixed_point_bwd(const float* lam, const float* rates,
                                 const int* cip, const int* ccols,
                                 const float* hist, float* dmu, float* dnb,
-                                float* dbusy, float* dlam, int E, int iters,
-                                int tid, int nt) {
+                                float* dbusy, float* dlam, int Eb,
+                                int Estride, int iters, int tid, int nt) {
     for (int t = iters; t >= 1; --t) {
-        const float* mu_prev = hist + (size_t)(t - 1) * E;
-        const float* mu_cur = hist + (size_t)t * E;
-        for (int e = tid; e < E; e += nt) {
+        const float* mu_prev = hist + (size_t)(t - 1) * Estride;
+        const float* mu_cur = hist + (size_t)t * Estride;
+        for (int e = tid; e < Eb; e += nt) {
             // mu_t = rates/(1+nb) => d nb = -mu_t^2/rates * d mu_t
             dnb[e] = -mu_cur[e] * mu_cur[e] / rates[e] * dmu[e];
         }
         __syncthreads();
-        for (int e = tid; e < E; e += nt) {
+        for (int e = tid; e < Eb; e += nt) {
             // busy feeds the nb of every conflicting link (A symmetric)
             float acc = 0.0f;
             for (int a = cip[e]; a < cip[e + 1]; ++a) acc += dnb[ccols[a]];
             dbusy[e] = acc;
         }
         __syncthreads();
-        for (int e = tid; e < E; e += nt) {
+        for (int e = tid; e < Eb; e += nt) {
             const float ratio = lam[e] / mu_prev[e];
             const float pass = (ratio <= 1.0f) ? 1.0f : 0.0f;  // ratio>=0
             const float g = pass * dbusy[e];
@@ -123,6 +125,7 @@ __global__ void critic_kernel(
     const int* __restrict__ conf_cols,     // flat local
     const float* __restrict__ rates,       // (B,E)
     const float* __restrict__ bw_comp,     // (B,C)
+    const int* __restrict__ E_arr,         // (B) real link counts
     float* __restrict__ grad_edge,         // (B,Ee) out (prezeroed)
     float* __restrict__ loss_out,          // (B,) out
     const float* __restrict__ T_arr,       // (B)
@@ -168,6 +171,7 @@ __global__ void critic_kernel(
     const float* ratesb = rates + (size_t)b * E;
     const float* bwb = bw_comp + (size_t)b * C;
     const float T = T_arr[b];
+    const int Eb = E_arr[b];
 
     for (int e = tid; e < Ee; e += nt) {
         lam_e[e] = 0.f;
@@ -193,9 +197,10 @@ __global__ void critic_kernel(
     __syncthreads();
 
     // ---- fixed point + unit delays ---------------------------------------
-    fixed_point_fwd(lam_e, ratesb, cip, ccols, hist, s1, E, iters, tid, nt);
+    fixed_point_fwd(lam_e, ratesb, cip, ccols, hist, s1, Eb, E, iters,
+                    tid, nt);
     const float* mu_last = hist + (size_t)iters * E;
-    for (int e = tid; e < E; e += nt)
+    for (int e = tid; e < Eb; e += nt)
         unit[e] = unit_fwd(lam_e[e], mu_last[e], T, 101.0f, cap);
     for (int k = tid; k < C; k += nt)
         unit[E + k] = unit_fwd(lam_e[E + k], bwb[k], T, 100.0f, cap);
@@ -223,7 +228,7 @@ __global__ void critic_kernel(
     __syncthreads();
 
     // ---- reverse: dunit → (dlam over links via fixed point, direct nodes)
-    for (int e = tid; e < E; e += nt) {
+    for (int e = tid; e < Eb; e += nt) {
         float dl_, dm_;
         unit_bwd(lam_e[e], mu_last[e], T, 101.0f, cap, dunit[e], &dl_, &dm_);
         dlam[e] += dl_;
@@ -236,7 +241,7 @@ __global__ void critic_kernel(
     }
     __syncthreads();
     fixed_point_bwd(lam_e, ratesb, cip, ccols, hist, s2 /*dmu*/, s1 /*dnb*/,
-                    s3 /*dbusy*/, dlam, E, iters, tid, nt);
+                    s3 /*dbusy*/, dlam, Eb, E, iters, tid, nt);
     __syncthreads();
 
     // ---- grad_routes prefix scan → grad_edge -----------------------------
@@ -309,11 +314,12 @@ __global__ void actor_head_fwd_kernel(
     float* dmb = dm + (size_t)b * N * N;
     const float T = T_arr[b];
 
-    for (int e = tid; e < E; e += nt) lam[e] = le[e];
-    __syncthreads();
-    fixed_point_fwd(lam, ratesb, cip, ccols, hist, busy, E, iters, tid, nt);
-    const float* mu_last = hist + (size_t)iters * E;
     const int Eb = E_arr[b];
+    for (int e = tid; e < Eb; e += nt) lam[e] = le[e];
+    __syncthreads();
+    fixed_point_fwd(lam, ratesb, cip, ccols, hist, busy, Eb, E, iters,
+                    tid, nt);
+    const float* mu_last = hist + (size_t)iters * E;
     for (int e = tid; e < Eb; e += nt) {
         const float d = unit_fwd(lam[e], mu_last[e], T, 101.0f, cap);
         const int u = edg[e * 2], v = edg[e * 2 + 1];
@@ -378,7 +384,8 @@ __global__ void actor_head_bwd_kernel(
     float* out = dlam_ext + (size_t)b * Ee;
     const float T = T_arr[b];
 
-    for (int e = tid; e < E; e += nt) lam[e] = le[e];
+    const int Eb = E_arr[b];
+    for (int e = tid; e < Eb; e += nt) lam[e] = le[e];
     if (!large) {
         float* h = lam + E;
         for (size_t i = tid; i < (size_t)(iters + 1) * E; i += nt)
@@ -387,15 +394,10 @@ __global__ void actor_head_bwd_kernel(
     __syncthreads();
     const float* mu_last = hist + (size_t)iters * E;
 
-    // link part: cotangent = gd[u,v] + gd[v,u] (dm wrote both);
-    // padded link slots [Eb, E) carry zero cotangent
-    const int Eb = E_arr[b];
-    for (int e = tid; e < E; e += nt) {
-        float dd = 0.f;
-        if (e < Eb) {
-            const int u = edg[e * 2], v = edg[e * 2 + 1];
-            dd = gd[(size_t)u * N + v] + gd[(size_t)v * N + u];
-        }
+    // link part: cotangent = gd[u,v] + gd[v,u] (dm wrote both)
+    for (int e = tid; e < Eb; e += nt) {
+        const int u = edg[e * 2], v = edg[e * 2 + 1];
+        const float dd = gd[(size_t)u * N + v] + gd[(size_t)v * N + u];
         float dl_, dm_;
         unit_bwd(lam[e], mu_last[e], T, 101.0f, cap, dd, &dl_, &dm_);
         dlam[e] = dl_;
@@ -403,9 +405,9 @@ __global__ void actor_head_bwd_kernel(
     }
     __syncthreads();
     fixed_point_bwd(lam, ratesb, cip, ccols, hist, dmu, s1, s2, dlam,
-                    E, iters, tid, nt);
+                    Eb, E, iters, tid, nt);
 
-    for (int e = tid; e < E; e += nt) out[e] = dlam[e];
+    for (int e = tid; e < E; e += nt) out[e] = e < Eb ? dlam[e] : 0.f;
     // node part: diagonal cotangent, direct (no fixed point)
     for (int n = tid; n < N; n += nt) {
         const long ve = nv[n];
@@ -429,7 +431,8 @@ std::vector<torch::Tensor> critic_hip(
     torch::Tensor mask, torch::Tensor rate, torch::Tensor ul,
     torch::Tensor dl, torch::Tensor conf_indptr, torch::Tensor conf_base,
     torch::Tensor conf_cols, torch::Tensor rates, torch::Tensor bw_comp,
-    torch::Tensor T_arr, long Ee, long iters, double cap) {
+    torch::Tensor E_arr, torch::Tensor T_arr, long Ee, long iters,
+    double cap) {
     const int B = route_links.size(0), J = route_links.size(1);
     const int H = route_links.size(2);
     const int E = rates.size(1), C = bw_comp.size(1);
@@ -463,7 +466,7 @@ std::vector<torch::Tensor> critic_hip(
                        conf_indptr.data_ptr<int>(),
                        conf_base.data_ptr<long>(),
                        conf_cols.data_ptr<int>(), rates.data_ptr<float>(),
-                       bw_comp.data_ptr<float>(),
+                       bw_comp.data_ptr<float>(), E_arr.data_ptr<int>(),
                        grad_edge.data_ptr<float>(), loss.data_ptr<float>(),
                        T_arr.data_ptr<float>(),
                        g_hist.data_ptr<float>(), g_dunit.data_ptr<float>(),

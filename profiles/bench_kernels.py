@@ -129,8 +129,8 @@ def bench_config(name, nodes, batch, gtype, distinct, reps):
         jobs.mask, jobs.rates.contiguous(), jobs.ul.contiguous(),
         jobs.dl.contiguous(), eng.k_conf_indptr, eng.k_conf_base,
         eng.k_conf_cols, eng.link_rates.contiguous(),
-        eng.bw_comp.contiguous(), eng.T_arr.contiguous(), eng.Ee,
-        eng.fp_iters, 0.0), reps)
+        eng.bw_comp.contiguous(), eng.k_E_arr, eng.T_arr.contiguous(),
+        eng.Ee, eng.fp_iters, 0.0), reps)
 
     def full_step():
         jb = eng.sample_jobs(0.15, gen)
