@@ -137,6 +137,12 @@ def main(argv=None):
     ap.add_argument("--explore_decay", type=float, default=0.999)
     ap.add_argument("--seed", type=int, default=100)
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--dtype", type=str, default=None,
+                    choices=[None, "float32", "float64"],
+                    help="override the compute dtype (default: fp32 on "
+                         "GPU, fp64 on CPU).  CPU fp32 vs fp64 with "
+                         "identical seeds is the precision drift study "
+                         "(docs/KERNELS.md)")
     ap.add_argument("--training_set", type=str, default="BAT1000")
     ap.add_argument("--model_root", type=str, default="model")
     ap.add_argument("--save_every", type=int, default=500)
@@ -206,7 +212,9 @@ def main(argv=None):
 
     rank, world = dp.init_from_env()
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
-    dtype = torch.float32 if device.startswith("cuda") else torch.float64
+    dtype = (getattr(torch, args.dtype) if args.dtype
+             else (torch.float32 if device.startswith("cuda")
+                   else torch.float64))
 
     model = ChebConvStack(K=args.K, dtype=dtype, seed=args.seed)
     resumed = False
