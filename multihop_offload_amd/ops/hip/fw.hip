@@ -36,24 +36,29 @@ __global__ void fw_lds_kernel(T* __restrict__ d, int N,
     const int n = n_arr ? n_arr[blockIdx.x] : N;
     const int tid = threadIdx.x;
     const int nt = blockDim.x;
-    const int total = n * n;
-    for (int c = tid; c < total; c += nt) s[c] = D[(c / n) * N + (c % n)];
+    // LDS row stride padded to a float4 multiple with +inf columns: the
+    // vectorized relax path then covers EVERY n (min-plus is inert on inf
+    // pads), instead of degrading to the scalar path when n % 4 != 0
+    const int ns = (sizeof(T) == 4) ? ((n + 3) & ~3) : n;
+    for (int c = tid; c < n * ns; c += nt) {
+        const int r = c / ns, j = c % ns;
+        s[c] = j < n ? D[(size_t)r * N + j] : (T)INFINITY;
+    }
     __syncthreads();
-    // thread layout: each of 32 lanes per half-wave owns a 4-column group
-    // (vectorized when n % 4 == 0 and fp32), rows strided across the
-    // remaining threads — d[k][j..j+3] loads once per k, d[i][k] broadcasts.
+    // thread layout: each of 32 lanes per half-wave owns a 4-column group,
+    // rows strided across the remaining threads — d[k][j..j+3] loads once
+    // per k, d[i][k] broadcasts.
     const int i0 = tid >> 5;
     const int istep = nt >> 5;
-    const bool vec4 = (sizeof(T) == 4) && (n % 4 == 0);
-    if (vec4) {
+    if (sizeof(T) == 4) {
         for (int k = 0; k < n; ++k) {
             for (int j = (tid & 31) * 4; j < n; j += 128) {
                 float4 dkj = *reinterpret_cast<const float4*>(
-                    reinterpret_cast<const float*>(s) + k * n + j);
+                    reinterpret_cast<const float*>(s) + k * ns + j);
                 for (int i = i0; i < n; i += istep) {
                     const float dik =
-                        reinterpret_cast<const float*>(s)[i * n + k];
-                    float* row = reinterpret_cast<float*>(s) + i * n + j;
+                        reinterpret_cast<const float*>(s)[i * ns + k];
+                    float* row = reinterpret_cast<float*>(s) + i * ns + j;
                     float4 cur = *reinterpret_cast<float4*>(row);
                     cur.x = fminf(cur.x, dik + dkj.x);
                     cur.y = fminf(cur.y, dik + dkj.y);
@@ -66,17 +71,18 @@ __global__ void fw_lds_kernel(T* __restrict__ d, int N,
         }
     } else {
         for (int k = 0; k < n; ++k) {
-            for (int j = tid & 127; j < n; j += 128) {
-                const T dkj = s[k * n + j];
-                for (int i = (tid >> 7); i < n; i += (nt >> 7)) {
-                    const T alt = s[i * n + k] + dkj;
-                    if (alt < s[i * n + j]) s[i * n + j] = alt;
+            for (int j = tid & 31; j < n; j += 32) {
+                const T dkj = s[k * ns + j];
+                for (int i = i0; i < n; i += istep) {
+                    const T alt = s[i * ns + k] + dkj;
+                    if (alt < s[i * ns + j]) s[i * ns + j] = alt;
                 }
             }
             __syncthreads();
         }
     }
-    for (int c = tid; c < total; c += nt) D[(c / n) * N + (c % n)] = s[c];
+    for (int c = tid; c < n * n; c += nt)
+        D[(size_t)(c / n) * N + (c % n)] = s[(c / n) * ns + (c % n)];
 }
 
 // ---- tiled (global-memory) path for large N -------------------------------
@@ -302,9 +308,10 @@ torch::Tensor floyd_warshall_hip(torch::Tensor w,
     const int* np_ = n_arr.has_value() ? n_arr->data_ptr<int>() : nullptr;
 
     AT_DISPATCH_FLOATING_TYPES(d.scalar_type(), "fw", [&] {
-        const size_t lds = (size_t)N * N * sizeof(scalar_t);
+        const int Ns = sizeof(scalar_t) == 4 ? ((N + 3) & ~3) : N;
+        const size_t lds = (size_t)N * Ns * sizeof(scalar_t);
         if (lds <= LDS_BYTES) {
-            hipLaunchKernelGGL(fw_lds_kernel<scalar_t>, dim3(B), dim3(256),
+            hipLaunchKernelGGL(fw_lds_kernel<scalar_t>, dim3(B), dim3(512),
                                lds, stream.stream(),
                                d.data_ptr<scalar_t>(), N, np_);
         } else {
