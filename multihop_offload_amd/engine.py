@@ -872,7 +872,8 @@ class EpisodeEngine:
     def gnn_episode(self, jobs: JobBatch, explore: float = 0.0,
                     gen: Optional[torch.Generator] = None,
                     train: bool = True, prob: bool = False,
-                    per_sample: bool = False) -> EpisodeResult:
+                    per_sample: bool = False,
+                    refine: int = 0) -> EpisodeResult:
         """One full GNN episode over the batch.  With ``train=True`` the
         summed per-instance actor gradients are left in ``model.param.grad``
         (caller applies the optimizer / DP all-reduce).  With
@@ -888,6 +889,20 @@ class EpisodeEngine:
         dst, est = self.offload_decide(jobs, sp, uds, explore, gen, prob)
         route_links, nhop, delay_emp, unit_mtx, written = \
             self._episode_eval(jobs, dst, sp)
+        # congestion-aware refinement (inference-time, opt-in): jobs whose
+        # analytic delay exceeds the horizon fall back to local compute and
+        # the assignment is re-evaluated.  Sound because job sources are
+        # distinct mobiles — a fallback only sheds load from links/servers
+        # (remaining delays can only improve) and adds load to the job's
+        # OWN processor (exactly the `local` method's delay).  Not part of
+        # the reference semantics; records produced with it say so.
+        for _ in range(int(refine) if not train else 0):
+            over = jobs.mask & (delay_emp > self.T_arr[:, None])
+            if not bool(over.any()):
+                break
+            dst = torch.where(over, jobs.sources, dst)
+            route_links, nhop, delay_emp, unit_mtx, written = \
+                self._episode_eval(jobs, dst, sp)
 
         loss_fn = loss_mse = None
         self.last_per_sample_grads = None

