@@ -296,6 +296,166 @@ __global__ void fw_phase3(T* __restrict__ d, int N, int nb, int kb) {
     }
 }
 
+// ---- TILE=64 fp32 specialization for large N (ER-1000 class) -------------
+// Halves the memory-side traffic of the TILE=32 path (4·N³/TILE bytes per
+// full FW) and gives each 256-thread block 8× the work per tile triple.
+constexpr int T64 = 64;
+
+__global__ void fw64_phase1(float* __restrict__ d, int N, int nb, int kb) {
+    __shared__ __attribute__((aligned(16))) float tile[T64 * T64];
+    float* D = d + (size_t)blockIdx.z * N * N;
+    const int tid = threadIdx.x;                     // 256 threads
+    for (int c = tid; c < T64 * T64; c += 256) {
+        const int r = kb * T64 + c / T64, cc = kb * T64 + c % T64;
+        tile[c] = (r < N && cc < N) ? D[(size_t)r * N + cc] : INFINITY;
+    }
+    __syncthreads();
+    // each thread owns 16 cells (row group of 4 × col group of 4)
+    const int ty = tid >> 4, tx = tid & 15;          // 16×16 thread grid
+    for (int k = 0; k < T64; ++k) {
+        float kj[4], ik[4];
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            kj[q] = tile[k * T64 + tx * 4 + q];
+            ik[q] = tile[(ty * 4 + q) * T64 + k];
+        }
+        __syncthreads();
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+#pragma unroll
+            for (int p = 0; p < 4; ++p) {
+                float* cell = tile + (ty * 4 + q) * T64 + tx * 4 + p;
+                const float alt = ik[q] + kj[p];
+                if (alt < *cell) *cell = alt;
+            }
+        __syncthreads();
+    }
+    for (int c = tid; c < T64 * T64; c += 256) {
+        const int r = kb * T64 + c / T64, cc = kb * T64 + c % T64;
+        if (r < N && cc < N) D[(size_t)r * N + cc] = tile[c];
+    }
+}
+
+__global__ void fw64_phase2(float* __restrict__ d, int N, int nb, int kb) {
+    __shared__ __attribute__((aligned(16))) float piv[T64 * T64];
+    __shared__ __attribute__((aligned(16))) float cur[T64 * T64];
+    float* D = d + (size_t)blockIdx.z * N * N;
+    int jb = blockIdx.x;
+    if (jb >= kb) jb += 1;
+    if (jb >= nb) return;
+    const int tid = threadIdx.x;
+    const bool row_strip = blockIdx.y == 0;
+    for (int c = tid; c < T64 * T64; c += 256) {
+        const int pr = kb * T64 + c / T64, pc = kb * T64 + c % T64;
+        piv[c] = (pr < N && pc < N) ? D[(size_t)pr * N + pc] : INFINITY;
+        int r, cc;
+        if (row_strip) { r = kb * T64 + c / T64; cc = jb * T64 + c % T64; }
+        else           { r = jb * T64 + c / T64; cc = kb * T64 + c % T64; }
+        cur[c] = (r < N && cc < N) ? D[(size_t)r * N + cc] : INFINITY;
+    }
+    __syncthreads();
+    const int ty = tid >> 4, tx = tid & 15;
+    if (row_strip) {
+        for (int k = 0; k < T64; ++k) {
+            float ik[4], kj[4];
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                ik[q] = piv[(ty * 4 + q) * T64 + k];
+                kj[q] = cur[k * T64 + tx * 4 + q];
+            }
+            __syncthreads();
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+#pragma unroll
+                for (int p = 0; p < 4; ++p) {
+                    float* cell = cur + (ty * 4 + q) * T64 + tx * 4 + p;
+                    const float alt = ik[q] + kj[p];
+                    if (alt < *cell) *cell = alt;
+                }
+            __syncthreads();
+        }
+    } else {
+        for (int k = 0; k < T64; ++k) {
+            float ik[4], kj[4];
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                ik[q] = cur[(ty * 4 + q) * T64 + k];
+                kj[q] = piv[k * T64 + tx * 4 + q];
+            }
+            __syncthreads();
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+#pragma unroll
+                for (int p = 0; p < 4; ++p) {
+                    float* cell = cur + (ty * 4 + q) * T64 + tx * 4 + p;
+                    const float alt = ik[q] + kj[p];
+                    if (alt < *cell) *cell = alt;
+                }
+            __syncthreads();
+        }
+    }
+    for (int c = tid; c < T64 * T64; c += 256) {
+        int r, cc;
+        if (row_strip) { r = kb * T64 + c / T64; cc = jb * T64 + c % T64; }
+        else           { r = jb * T64 + c / T64; cc = kb * T64 + c % T64; }
+        if (r < N && cc < N) D[(size_t)r * N + cc] = cur[c];
+    }
+}
+
+// phase 3: 256 threads, 4×4 registers per thread over a 64×64 output tile;
+// independent min-add chains, no barriers inside the k-loop.
+__global__ void fw64_phase3(float* __restrict__ d, int N, int nb, int kb) {
+    __shared__ __attribute__((aligned(16))) float rowt[T64 * T64];
+    __shared__ __attribute__((aligned(16))) float colt[T64 * T64];
+    float* D = d + (size_t)blockIdx.z * N * N;
+    int ib = blockIdx.y, jb = blockIdx.x;
+    if (ib >= kb) ib += 1;
+    if (jb >= kb) jb += 1;
+    if (ib >= nb || jb >= nb) return;
+    const int tid = threadIdx.x;
+    for (int c = tid; c < T64 * T64; c += 256) {
+        int r = ib * T64 + c / T64, cc = kb * T64 + c % T64;
+        rowt[c] = (r < N && cc < N) ? D[(size_t)r * N + cc] : INFINITY;
+        r = kb * T64 + c / T64;
+        cc = jb * T64 + c % T64;
+        colt[c] = (r < N && cc < N) ? D[(size_t)r * N + cc] : INFINITY;
+    }
+    __syncthreads();
+    const int ty = tid >> 4, tx = tid & 15;
+    const int r0 = ib * T64 + ty * 4;
+    const int c0 = jb * T64 + tx * 4;
+    float4 v[4];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+        v[q] = {INFINITY, INFINITY, INFINITY, INFINITY};
+        if (r0 + q < N)
+#pragma unroll
+            for (int p = 0; p < 4; ++p)
+                if (c0 + p < N)
+                    (&v[q].x)[p] = D[(size_t)(r0 + q) * N + c0 + p];
+    }
+#pragma unroll 2
+    for (int k = 0; k < T64; ++k) {
+        const float4 ckj =
+            *reinterpret_cast<const float4*>(colt + k * T64 + tx * 4);
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            const float dik = rowt[(ty * 4 + q) * T64 + k];
+            v[q].x = fminf(v[q].x, dik + ckj.x);
+            v[q].y = fminf(v[q].y, dik + ckj.y);
+            v[q].z = fminf(v[q].z, dik + ckj.z);
+            v[q].w = fminf(v[q].w, dik + ckj.w);
+        }
+    }
+#pragma unroll
+    for (int q = 0; q < 4; ++q)
+        if (r0 + q < N)
+#pragma unroll
+            for (int p = 0; p < 4; ++p)
+                if (c0 + p < N)
+                    D[(size_t)(r0 + q) * N + c0 + p] = (&v[q].x)[p];
+}
+
 }  // namespace
 
 torch::Tensor floyd_warshall_hip(torch::Tensor w,
@@ -314,6 +474,24 @@ torch::Tensor floyd_warshall_hip(torch::Tensor w,
             hipLaunchKernelGGL(fw_lds_kernel<scalar_t>, dim3(B), dim3(512),
                                lds, stream.stream(),
                                d.data_ptr<scalar_t>(), N, np_);
+        } else if constexpr (std::is_same_v<scalar_t, float>) {
+            // fp32 large-N: TILE=64 (half the fabric traffic of TILE=32,
+            // 8x the work per block; ER-1000 class)
+            const int nb = (N + T64 - 1) / T64;
+            for (int kb = 0; kb < nb; ++kb) {
+                hipLaunchKernelGGL(fw64_phase1, dim3(1, 1, B), dim3(256), 0,
+                                   stream.stream(), d.data_ptr<float>(),
+                                   N, nb, kb);
+                if (nb > 1) {
+                    hipLaunchKernelGGL(fw64_phase2, dim3(nb - 1, 2, B),
+                                       dim3(256), 0, stream.stream(),
+                                       d.data_ptr<float>(), N, nb, kb);
+                    hipLaunchKernelGGL(fw64_phase3,
+                                       dim3(nb - 1, nb - 1, B), dim3(256),
+                                       0, stream.stream(),
+                                       d.data_ptr<float>(), N, nb, kb);
+                }
+            }
         } else {
             const int nb = (N + TILE - 1) / TILE;
             dim3 thr(TILE, TILE);
@@ -326,18 +504,11 @@ torch::Tensor floyd_warshall_hip(torch::Tensor w,
                                        dim3(nb - 1, 2, B), thr, 0,
                                        stream.stream(),
                                        d.data_ptr<scalar_t>(), N, nb, kb);
-                    if constexpr (std::is_same_v<scalar_t, float>) {
-                        hipLaunchKernelGGL(fw_phase3_f32,
-                                           dim3(nb - 1, nb - 1, B),
-                                           dim3(8, 8), 0, stream.stream(),
-                                           d.data_ptr<float>(), N, nb, kb);
-                    } else {
-                        hipLaunchKernelGGL(fw_phase3<scalar_t>,
-                                           dim3(nb - 1, nb - 1, B), thr, 0,
-                                           stream.stream(),
-                                           d.data_ptr<scalar_t>(), N, nb,
-                                           kb);
-                    }
+                    hipLaunchKernelGGL(fw_phase3<scalar_t>,
+                                       dim3(nb - 1, nb - 1, B), thr, 0,
+                                       stream.stream(),
+                                       d.data_ptr<scalar_t>(), N, nb,
+                                       kb);
                 }
             }
         }
