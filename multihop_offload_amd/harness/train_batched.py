@@ -305,6 +305,8 @@ def main(argv=None):
     best_eval_params = None
     best_eval_opt = None
     eval_rollbacks = 0
+    eval_hist = []
+    last_rollback_step = -10**9
     loads = ([float(x) for x in args.arrival_scales.split(",")]
              if args.arrival_scales else [args.arrival_scale])
     profiler = None
@@ -410,25 +412,56 @@ def main(argv=None):
             eval_congest = _dist_mean(ec, world, engine.device)
             eval_metric = eval_tau + args.eval_congest_weight * eval_congest
             rolled = 0
+            # dead-policy rescue: a collapsed run can freeze in a
+            # zero-output attractor (dead output ReLU → zero gradients →
+            # bit-identical eval forever, observed at seed-dependent
+            # rates).  Three identical evals ⇒ re-wake the output bias
+            # and re-explore.
+            eval_hist.append(eval_metric)
+            dead = (len(eval_hist) >= 3
+                    and max(eval_hist[-3:]) - min(eval_hist[-3:]) < 1e-9)
             if eval_metric < best_eval_tau:
                 best_eval_tau = eval_metric
                 best_eval_params = [p.detach().clone()
                                     for p in engine.model.parameters()]
                 best_eval_opt = _copy.deepcopy(opt.state_dict())
-            elif (args.eval_guard and best_eval_params is not None
-                  and eval_metric > args.eval_guard
-                  * max(best_eval_tau, 20.0)):
-                # held-out collapse: restore the best-eval state and cool
-                # the lr — turns the bistable divergence into a recoverable
-                # excursion instead of a lost run
-                with torch.no_grad():
-                    for p, bp in zip(engine.model.parameters(),
-                                     best_eval_params):
-                        p.copy_(bp)
-                if best_eval_opt is not None:
-                    opt.load_state_dict(best_eval_opt)
+            elif dead or (args.eval_guard and best_eval_params is not None
+                          and eval_metric > args.eval_guard
+                          * max(best_eval_tau, 20.0)):
+                # held-out collapse: restore the best-eval state (keep the
+                # SCHEDULE lr — restoring the snapshot's lr would undo
+                # decays), re-explore, and cool the lr only when collapses
+                # repeat back-to-back (a single excursion self-recovers;
+                # compounding cuts freeze the run at its early best)
+                cur_lr = opt.param_groups[0]["lr"]
+                if best_eval_params is not None:
+                    with torch.no_grad():
+                        for p, bp in zip(engine.model.parameters(),
+                                         best_eval_params):
+                            p.copy_(bp)
+                    if best_eval_opt is not None:
+                        opt.load_state_dict(best_eval_opt)
+                if dead:
+                    # frozen policy (possibly the restored best itself, if
+                    # the run died at init): re-wake the output layer
+                    with torch.no_grad():
+                        engine.model.layers[-1].bias.fill_(0.5)
+                        if best_eval_params is not None and \
+                                best_eval_tau >= 200.0:
+                            # the "best" is the dead policy — drop it so a
+                            # woken policy can take over
+                            best_eval_tau = float("inf")
+                            best_eval_params = None
+                            best_eval_opt = None
+                repeated = (step - last_rollback_step
+                            <= 3 * args.eval_every)
                 for group in opt.param_groups:
-                    group["lr"] /= 3.0
+                    group["lr"] = cur_lr / (3.0 if repeated else 1.0)
+                explore = max(explore, 0.05)
+                if explore_dev is not None:
+                    explore_dev.fill_(max(float(explore_dev), 0.05))
+                last_rollback_step = step
+                eval_hist.clear()
                 hip_graphs.clear()
                 eval_rollbacks += 1
                 rolled = 1
