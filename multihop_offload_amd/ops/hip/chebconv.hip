@@ -62,6 +62,40 @@ DEV_INLINE void gemm_acc(const float* __restrict__ src,
     }
 }
 
+// ---- weight-gradient tile, A-operand straight from GLOBAL memory ---------
+// dW[i][j] = sum_r g[r][i] * delta[r][j]; g is the (Ee,32) row-major saved
+// activation in HBM/L3 — each element is read exactly once per wgrad, so
+// staging it through LDS first (two extra sweeps + barriers per layer)
+// buys nothing; the 8 independent tile-chains per wave cover the load
+// latency.
+DEV_INLINE void gemm_wgrad_g(const float* __restrict__ g, int Ee,
+                             const float* __restrict__ delta,
+                             float* __restrict__ dw_out,  // global [32][32]
+                             int rows_pad, int tid) {
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int nw = blockDim.x >> 6;
+    const int k_in = lane >> 4;
+    const int c_in = lane & 15;
+    for (int t = wid; t < 8; t += nw) {
+        const int tile = t >> 1, half = t & 1;
+        const int i0 = (tile >> 1) * 16, j0 = (tile & 1) * 16;
+        const int kk0 = half * (rows_pad / 8);
+        const int kk1 = kk0 + rows_pad / 8;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        for (int kk = kk0; kk < kk1; ++kk) {
+            const int r = kk * 4 + k_in;
+            const float a = r < Ee ? g[(size_t)r * F + i0 + c_in] : 0.f;
+            const float b = delta[r * STRIDE + j0 + c_in];
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            atomicAdd(&dw_out[(i0 + (lane >> 4) * 4 + r) * F + j0
+                              + (lane & 15)], acc[r]);
+    }
+}
+
 // ---- weight-gradient tile: dW[i][j] = sum_r src[r][i] * delta[r][j] ------
 // MFMA over the row (K) dimension: 2×2 tiles of 16×16, one per wave.
 DEV_INLINE void gemm_wgrad(const float* __restrict__ src,
@@ -422,44 +456,51 @@ __global__ void cheb_bwd_kernel(
 
     for (int l = L - 1; l >= 0; --l) {
         const bool last = (l == L - 1);
-        // activation mask from stored post-act X_{l+1} (coalesced float4)
+        // activation mask from stored post-act X_{l+1} (coalesced float4),
+        // with the bias gradient folded into the same sweep
         const float slope = last ? 0.f : 0.2f;
-        for (int t = tid; t < Ee * (F / 4); t += nt) {
-            const int r = t >> 3;
-            const int c = (t & 7) * 4;
-            const float4 v = *reinterpret_cast<const float4*>(
-                actsb + ((size_t)(l + 1) * Ee + r) * F + c);
-            float* d = Db + r * STRIDE + c;
-            d[0] *= v.x > 0.f ? 1.f : slope;
-            d[1] *= v.y > 0.f ? 1.f : slope;
-            d[2] *= v.z > 0.f ? 1.f : slope;
-            d[3] *= v.w > 0.f ? 1.f : slope;
+        {
+            float dbacc[4] = {0.f, 0.f, 0.f, 0.f};
+            int lastc = -1;
+            for (int t = tid; t < Ee * (F / 4); t += nt) {
+                const int r = t >> 3;
+                const int c = (t & 7) * 4;
+                const float4 v = *reinterpret_cast<const float4*>(
+                    actsb + ((size_t)(l + 1) * Ee + r) * F + c);
+                float* d = Db + r * STRIDE + c;
+                d[0] *= v.x > 0.f ? 1.f : slope;
+                d[1] *= v.y > 0.f ? 1.f : slope;
+                d[2] *= v.z > 0.f ? 1.f : slope;
+                d[3] *= v.w > 0.f ? 1.f : slope;
+                if (c != lastc && lastc >= 0) {
+                    // column group changed: flush the partial db sums
+#pragma unroll
+                    for (int q = 0; q < 4; ++q) {
+                        atomicAdd(&dbb[l * F + lastc + q], dbacc[q]);
+                        dbacc[q] = 0.f;
+                    }
+                }
+                lastc = c;
+                dbacc[0] += d[0]; dbacc[1] += d[1];
+                dbacc[2] += d[2]; dbacc[3] += d[3];
+            }
+            if (lastc >= 0)
+#pragma unroll
+                for (int q = 0; q < 4; ++q)
+                    atomicAdd(&dbb[l * F + lastc + q], dbacc[q]);
         }
-        // load X_l
-        load_acts(Ab, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
         for (int i = tid; i < K * F * F; i += nt)
             Wl[i] = W[((size_t)l * K) * F * F + i];
         __syncthreads();
 
-        // db[j] = sum_r Db[r][j] — (j, row-chunk) threads, db prezeroed
-        {
-            const int nchunk = nt / F;
-            const int j = tid % F, ch = tid / F;
-            float acc = 0.f;
-            for (int r = ch; r < Ee; r += nchunk) acc += Db[r * STRIDE + j];
-            atomicAdd(&dbb[l * F + j], acc);
-        }
-        gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
-        if (K > 1) {
-            __syncthreads();
-            load_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee, rows_pad,
-                      tid, nt);                           // T1 from forward
-            __syncthreads();
-            gemm_wgrad(Tb, Db, dWb + ((size_t)l * K + 1) * F * F, rows_pad,
-                       tid);
-        }
+        // weight gradients straight from the saved global activations —
+        // no LDS staging sweeps (each element is consumed exactly once)
+        gemm_wgrad_g(actsb + (size_t)l * Ee * F, Ee, Db,
+                     dWb + ((size_t)l * K) * F * F, rows_pad, tid);
+        if (K > 1)
+            gemm_wgrad_g(t1s + ((size_t)b * L + l) * Ee * F, Ee, Db,
+                         dWb + ((size_t)l * K + 1) * F * F, rows_pad, tid);
         if (l == 0) break;                       // features are leaves
-        __syncthreads();
         // U = Db·W1ᵀ (into Tb) and dX = Db·W0ᵀ (into Ab), fused per tile
         gemm_dx_fused(Db, Ab, Tb, Wl, K, rows_pad, tid);
         __syncthreads();
@@ -467,8 +508,7 @@ __global__ void cheb_bwd_kernel(
             spmv(Tb, Ab, ipt, cls, Ee, rows_pad, tid, nt, 1); // dX += A·U
             __syncthreads();
         }
-        // swap: Ab (dX) becomes the delta of the layer below; the old
-        // delta buffer is reused as the next X_l staging area
+        // swap: Ab (dX) becomes the delta of the layer below
         float* tmp = Ab;
         Ab = Db;
         Db = tmp;
