@@ -397,6 +397,9 @@ def main(argv=None):
                 for p in engine.model.parameters():
                     if p.grad is not None:
                         n = p.grad.norm().clamp(min=1e-12)
+                        if not torch.isfinite(n):
+                            p.grad.zero_()   # poisoned step: skip tensor
+                            continue
                         p.grad *= torch.clamp(n, max=1.0) / n
             opt.step()
             engine.model.apply_constraints()

@@ -57,18 +57,26 @@ __global__ void fused_adam_kernel(
         }
         if (tid == 0) {
             const float nrm = sqrtf(red[0]);
-            factor = nrm > 1.0f ? 1.0f / nrm : 1.0f;
+            // non-finite gradient norm (inf/NaN from the 1/(mu-lam) pole
+            // in fp32): SKIP this tensor's update entirely — one poisoned
+            // step would otherwise write NaN into p/m/v and freeze the
+            // policy permanently (observed at seed-dependent rates)
+            factor = isfinite(nrm) ? (nrm > 1.0f ? 1.0f / nrm : 1.0f)
+                                   : -1.0f;
         }
         __syncthreads();
         const float f = factor;
-        for (int i = tid; i < size; i += nt) {
-            const int j = off + i;
-            const float gv = g[j] * f;
-            const float mn = beta1 * m[j] + (1.f - beta1) * gv;
-            const float vn = beta2 * v[j] + (1.f - beta2) * gv * gv;
-            m[j] = mn;
-            v[j] = vn;
-            p[j] -= lr * (mn * bc1) / (sqrtf(vn * bc2) + eps);
+        if (f >= 0.f) {
+            for (int i = tid; i < size; i += nt) {
+                const int j = off + i;
+                float gv = g[j] * f;
+                gv = isfinite(gv) ? gv : 0.f;   // isolated inf/NaN lanes
+                const float mn = beta1 * m[j] + (1.f - beta1) * gv;
+                const float vn = beta2 * v[j] + (1.f - beta2) * gv * gv;
+                m[j] = mn;
+                v[j] = vn;
+                p[j] -= lr * (mn * bc1) / (sqrtf(vn * bc2) + eps);
+            }
         }
         __syncthreads();
     }
