@@ -493,14 +493,23 @@ __global__ void cheb_bwd_kernel(
             Wl[i] = W[((size_t)l * K) * F * F + i];
         __syncthreads();
 
-        // weight gradients straight from the saved global activations —
-        // no LDS staging sweeps (each element is consumed exactly once)
-        gemm_wgrad_g(actsb + (size_t)l * Ee * F, Ee, Db,
-                     dWb + ((size_t)l * K) * F * F, rows_pad, tid);
-        if (K > 1)
-            gemm_wgrad_g(t1s + ((size_t)b * L + l) * Ee * F, Ee, Db,
-                         dWb + ((size_t)l * K + 1) * F * F, rows_pad, tid);
+        // weight gradients from LDS-staged activations.  (Feeding the
+        // MFMA tiles straight from global was MEASURED 2x slower — the
+        // dependent-accumulator chain stalls on the ~400-cycle loads;
+        // profiles/r02_notes.md.)
+        load_acts(Ab, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
+        __syncthreads();
+        gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
+        if (K > 1) {
+            __syncthreads();
+            load_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee, rows_pad,
+                      tid, nt);                           // T1 from forward
+            __syncthreads();
+            gemm_wgrad(Tb, Db, dWb + ((size_t)l * K + 1) * F * F, rows_pad,
+                       tid);
+        }
         if (l == 0) break;                       // features are leaves
+        __syncthreads();
         // U = Db·W1ᵀ (into Tb) and dX = Db·W0ᵀ (into Ab), fused per tile
         gemm_dx_fused(Db, Ab, Tb, Wl, K, rows_pad, tid);
         __syncthreads();

@@ -29,7 +29,11 @@ def init_from_env() -> tuple[int, int]:
     dist.init_process_group(backend=backend)
     rank = dist.get_rank()
     if backend == "nccl":
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        # modulo the visible device count so an oversubscribed probe
+        # (2 ranks on 1 GPU — the multi-rank RCCL path exercised on a
+        # single-GPU lease) maps both ranks onto device 0
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank))
+                              % max(torch.cuda.device_count(), 1))
     return rank, dist.get_world_size()
 
 
