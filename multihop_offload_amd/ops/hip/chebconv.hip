@@ -456,49 +456,37 @@ __global__ void cheb_bwd_kernel(
 
     for (int l = L - 1; l >= 0; --l) {
         const bool last = (l == L - 1);
-        // activation mask from stored post-act X_{l+1} (coalesced float4),
-        // with the bias gradient folded into the same sweep
+        // activation mask from stored post-act X_{l+1} (coalesced float4)
         const float slope = last ? 0.f : 0.2f;
-        {
-            float dbacc[4] = {0.f, 0.f, 0.f, 0.f};
-            int lastc = -1;
-            for (int t = tid; t < Ee * (F / 4); t += nt) {
-                const int r = t >> 3;
-                const int c = (t & 7) * 4;
-                const float4 v = *reinterpret_cast<const float4*>(
-                    actsb + ((size_t)(l + 1) * Ee + r) * F + c);
-                float* d = Db + r * STRIDE + c;
-                d[0] *= v.x > 0.f ? 1.f : slope;
-                d[1] *= v.y > 0.f ? 1.f : slope;
-                d[2] *= v.z > 0.f ? 1.f : slope;
-                d[3] *= v.w > 0.f ? 1.f : slope;
-                if (c != lastc && lastc >= 0) {
-                    // column group changed: flush the partial db sums
-#pragma unroll
-                    for (int q = 0; q < 4; ++q) {
-                        atomicAdd(&dbb[l * F + lastc + q], dbacc[q]);
-                        dbacc[q] = 0.f;
-                    }
-                }
-                lastc = c;
-                dbacc[0] += d[0]; dbacc[1] += d[1];
-                dbacc[2] += d[2]; dbacc[3] += d[3];
-            }
-            if (lastc >= 0)
-#pragma unroll
-                for (int q = 0; q < 4; ++q)
-                    atomicAdd(&dbb[l * F + lastc + q], dbacc[q]);
+        for (int t = tid; t < Ee * (F / 4); t += nt) {
+            const int r = t >> 3;
+            const int c = (t & 7) * 4;
+            const float4 v = *reinterpret_cast<const float4*>(
+                actsb + ((size_t)(l + 1) * Ee + r) * F + c);
+            float* d = Db + r * STRIDE + c;
+            d[0] *= v.x > 0.f ? 1.f : slope;
+            d[1] *= v.y > 0.f ? 1.f : slope;
+            d[2] *= v.z > 0.f ? 1.f : slope;
+            d[3] *= v.w > 0.f ? 1.f : slope;
         }
+        // load X_l  (feeding the wgrad MFMA tiles straight from global
+        // was MEASURED 2x slower — dependent-accumulator stalls on the
+        // ~400-cycle loads; and folding db into the mask sweep above was
+        // measured +13% — atomic flush every column-group change.
+        // profiles/r02_notes.md: measure, don't guess.)
+        load_acts(Ab, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
         for (int i = tid; i < K * F * F; i += nt)
             Wl[i] = W[((size_t)l * K) * F * F + i];
         __syncthreads();
 
-        // weight gradients from LDS-staged activations.  (Feeding the
-        // MFMA tiles straight from global was MEASURED 2x slower — the
-        // dependent-accumulator chain stalls on the ~400-cycle loads;
-        // profiles/r02_notes.md.)
-        load_acts(Ab, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
-        __syncthreads();
+        // db[j] = sum_r Db[r][j] — (j, row-chunk) threads, db prezeroed
+        {
+            const int nchunk = nt / F;
+            const int j = tid % F, ch = tid / F;
+            float acc = 0.f;
+            for (int r = ch; r < Ee; r += nchunk) acc += Db[r * STRIDE + j];
+            atomicAdd(&dbb[l * F + j], acc);
+        }
         gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
         if (K > 1) {
             __syncthreads();
