@@ -223,7 +223,12 @@ __global__ void walk_eval_kernel(
     // ---- stage 1: walk + load accumulation -------------------------------
     for (int j = tid; j < J; j += nt) {
         const size_t bj = (size_t)b * J + j;
-        if (!mask[bj]) { nhop[bj] = 0; continue; }
+        if (!mask[bj]) {
+            nhop[bj] = 0;
+            for (int h = 0; h < H; ++h)
+                route_links[bj * H + h] = -1;
+            continue;
+        }
         const int d = (int)dstv[bj];
         int node = (int)src[bj];
         const float add = (ul[bj] + dl[bj]) * rate[bj];
@@ -244,6 +249,8 @@ __global__ void walk_eval_kernel(
         }
         if (node != d) atomicAdd(&overflow[b], 1);
         nhop[bj] = h;
+        for (int h2 = h; h2 < H; ++h2)       // outputs arrive torch::empty
+            route_links[bj * H + h2] = -1;
         atomicAdd(&sload[d], ul[bj] * rate[bj]);
     }
     __syncthreads();
@@ -305,13 +312,15 @@ __global__ void walk_eval_kernel(
     // ---- stage 3b: unpack last-job-wins units into unit_mtx/written ------
     float* um = unit_mtx + (size_t)b * N * N;
     bool* wm = written + (size_t)b * N * N;
-    for (int i = tid; i < nb_ * nb_; i += nt) {
-        const int cell = (i / nb_) * N + (i % nb_);
-        const unsigned long long pk = upk[cell];
-        if (pk) {
-            um[cell] = __uint_as_float((unsigned int)(pk & 0xffffffffull));
-            wm[cell] = true;
-        }
+    // outputs arrive torch::empty: every cell is written exactly once
+    // (pad region → 0/false), replacing two fill launches per call
+    for (int i = tid; i < N * N; i += nt) {
+        const int r = i / N, c = i % N;
+        const unsigned long long pk =
+            (r < nb_ && c < nb_) ? upk[i] : 0ull;
+        um[i] = pk ? __uint_as_float((unsigned int)(pk & 0xffffffffull))
+                   : 0.f;
+        wm[i] = pk != 0ull;
     }
 }
 
@@ -355,12 +364,12 @@ std::vector<torch::Tensor> walk_eval_hip(
     const int J = src.size(1), E = rates.size(1);
     auto opts_i = adj_indptr.options();
     auto opts_f = sp.options();
-    auto route_links = torch::full({B, J, H}, -1,
-                                   opts_i.dtype(torch::kInt32));
-    auto nhop = torch::zeros({B, J}, opts_i.dtype(torch::kInt32));
+    // all outputs fully initialized in-kernel (no fill launches)
+    auto route_links = torch::empty({B, J, H}, opts_i.dtype(torch::kInt32));
+    auto nhop = torch::empty({B, J}, opts_i.dtype(torch::kInt32));
     auto delay_emp = torch::empty({B, J}, opts_f);
-    auto unit_mtx = torch::zeros({B, N, N}, opts_f);
-    auto written = torch::zeros({B, N, N}, opts_f.dtype(torch::kBool));
+    auto unit_mtx = torch::empty({B, N, N}, opts_f);
+    auto written = torch::empty({B, N, N}, opts_f.dtype(torch::kBool));
     // u64 scratch for deterministic last-job-wins unit writes (the kernel
     // zeroes its own slice — empty, not zeros)
     auto upack = torch::empty({B, N, N}, opts_f.dtype(torch::kInt64));

@@ -436,12 +436,12 @@ std::vector<torch::Tensor> critic_hip(
     const int B = route_links.size(0), J = route_links.size(1);
     const int H = route_links.size(2);
     const int E = rates.size(1), C = bw_comp.size(1);
-    auto grad_edge = torch::zeros({B, (long)Ee}, rates.options());
-    auto loss = torch::zeros({B}, rates.options());
+    auto loss = torch::empty({B}, rates.options());
     size_t lds = sizeof(float) *
         (5 * (size_t)Ee + (size_t)(iters + 1) * E + 3 * (size_t)E);
     int large = 0;
     torch::Tensor g_hist, g_dunit, g_dlam;
+    torch::Tensor grad_edge;
     if (lds > 160 * 1024) {
         large = 1;
         lds = sizeof(float) * (2 * (size_t)Ee + 3 * (size_t)E);
@@ -450,10 +450,13 @@ std::vector<torch::Tensor> critic_hip(
         g_hist = torch::empty({B, iters + 1, (long)E}, rates.options());
         g_dunit = torch::zeros({B, (long)Ee}, rates.options());
         g_dlam = torch::zeros({B, (long)Ee}, rates.options());
+        grad_edge = torch::zeros({B, (long)Ee}, rates.options());
     } else {
         g_hist = torch::empty({1}, rates.options());
         g_dunit = g_hist;
         g_dlam = g_hist;
+        // small mode copies the full dgre LDS row out — no prezero needed
+        grad_edge = torch::empty({B, (long)Ee}, rates.options());
     }
     const int threads = E >= 1500 ? 1024 : 256;
     auto stream = at::cuda::getCurrentCUDAStream();
@@ -483,7 +486,10 @@ std::vector<torch::Tensor> actor_head_fwd_hip(
     double cap) {
     const int B = lam_ext.size(0), Ee = lam_ext.size(1);
     const int E = rates.size(1), C = bw_comp.size(1);
-    auto dm = torch::zeros({B, N, N}, lam_ext.options());
+    // every cell a consumer reads (real links both directions + the full
+    // diagonal) is written by the kernel; non-edge cells are masked by
+    // adj / `written` everywhere downstream — skip the fill launch
+    auto dm = torch::empty({B, N, N}, lam_ext.options());
     auto mu_hist = torch::empty({B, iters + 1, (long)E}, lam_ext.options());
     size_t lds = sizeof(float) * ((size_t)(iters + 2) * E + E);
     int large = 0;
