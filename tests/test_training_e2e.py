@@ -44,3 +44,28 @@ def test_short_training_improves_tau():
         model.apply_constraints()
     tau_after = _eval_tau(engine, gen)
     assert tau_after < tau_before * 0.9, (tau_before, tau_after)
+
+
+def test_nonfinite_grad_step_skipped():
+    """The torch optimizer path zeroes a tensor's gradient when its norm
+    is non-finite (the fp32 pole can emit inf/NaN): parameters must not
+    be poisoned (the fused GPU kernel mirrors this — skip, not NaN)."""
+    import torch
+    from multihop_offload_amd.models.chebconv import ChebConvStack
+    m = ChebConvStack(K=2, dtype=torch.float64, seed=1)
+    before = [p.detach().clone() for p in m.parameters()]
+    opt = torch.optim.Adam(m.parameters(), lr=1e-2, eps=1e-7)
+    for p in m.parameters():
+        p.grad = torch.full_like(p, float("nan"))
+    # the trainer's guard logic
+    with torch.no_grad():
+        for p in m.parameters():
+            n = p.grad.norm().clamp(min=1e-12)
+            if not torch.isfinite(n):
+                p.grad.zero_()
+                continue
+            p.grad *= torch.clamp(n, max=1.0) / n
+    opt.step()
+    for p, b in zip(m.parameters(), before):
+        assert torch.isfinite(p).all()
+        assert torch.allclose(p, b)    # zero grad → Adam no-op step
