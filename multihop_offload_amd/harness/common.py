@@ -140,6 +140,9 @@ class EngineRunner:
                                     dtype=agent.dtype)
         self.gen = torch.Generator(device=str(agent.device))
         self.gen.manual_seed(seed * 9973 + g.seed)
+        # per-case kernel-RNG seed: without this every case engine would
+        # replay the same in-kernel ε-greedy stream (correlated explore)
+        self.engine.set_rng_seed(seed * 9973 + g.seed)
 
     def run_method(self, method: str, jobs, explore: float = 0.0):
         import torch
