@@ -62,40 +62,6 @@ DEV_INLINE void gemm_acc(const float* __restrict__ src,
     }
 }
 
-// ---- weight-gradient tile, A-operand straight from GLOBAL memory ---------
-// dW[i][j] = sum_r g[r][i] * delta[r][j]; g is the (Ee,32) row-major saved
-// activation in HBM/L3 — each element is read exactly once per wgrad, so
-// staging it through LDS first (two extra sweeps + barriers per layer)
-// buys nothing; the 8 independent tile-chains per wave cover the load
-// latency.
-DEV_INLINE void gemm_wgrad_g(const float* __restrict__ g, int Ee,
-                             const float* __restrict__ delta,
-                             float* __restrict__ dw_out,  // global [32][32]
-                             int rows_pad, int tid) {
-    const int lane = tid & 63;
-    const int wid = tid >> 6;
-    const int nw = blockDim.x >> 6;
-    const int k_in = lane >> 4;
-    const int c_in = lane & 15;
-    for (int t = wid; t < 8; t += nw) {
-        const int tile = t >> 1, half = t & 1;
-        const int i0 = (tile >> 1) * 16, j0 = (tile & 1) * 16;
-        const int kk0 = half * (rows_pad / 8);
-        const int kk1 = kk0 + rows_pad / 8;
-        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-        for (int kk = kk0; kk < kk1; ++kk) {
-            const int r = kk * 4 + k_in;
-            const float a = r < Ee ? g[(size_t)r * F + i0 + c_in] : 0.f;
-            const float b = delta[r * STRIDE + j0 + c_in];
-            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
-        }
-#pragma unroll
-        for (int r = 0; r < 4; ++r)
-            atomicAdd(&dw_out[(i0 + (lane >> 4) * 4 + r) * F + j0
-                              + (lane & 15)], acc[r]);
-    }
-}
-
 // ---- weight-gradient tile: dW[i][j] = sum_r src[r][i] * delta[r][j] ------
 // MFMA over the row (K) dimension: 2×2 tiles of 16×16, one per wave.
 DEV_INLINE void gemm_wgrad(const float* __restrict__ src,

@@ -153,122 +153,6 @@ __global__ void fw_phase2(T* __restrict__ d, int N, int nb, int kb) {
     if (r < N && c < N) D[r * N + c] = cur[ty * TILE + tx];
 }
 
-// float specialization of phase 3: register-blocked 4×4 cells per thread
-// (one 64-thread wave per 32×32 tile): per k, 4 scalar + 1 float4 LDS reads
-// feed 16 min-add chains.  Launch dim3(8, 8).
-__global__ void fw_phase3_f32(float* __restrict__ d, int N, int nb, int kb) {
-    __shared__ __attribute__((aligned(16))) float rowt[TILE * TILE];
-    __shared__ __attribute__((aligned(16))) float colt[TILE * TILE];
-    float* D = d + (size_t)blockIdx.z * N * N;
-    int ib = blockIdx.y, jb = blockIdx.x;
-    if (ib >= kb) ib += 1;
-    if (jb >= kb) jb += 1;
-    if (ib >= nb || jb >= nb) return;
-    const int tx = threadIdx.x;              // 0..7 → 4 columns
-    const int ty = threadIdx.y;              // 0..7 → 4 rows
-    const int tid = ty * 8 + tx;
-    // cooperative tile loads: 64 threads × 16 floats per tile
-    {
-        const int lr = tid >> 1;             // 0..31
-        const int lc = (tid & 1) * 16;       // two float4x4 spans per row
-        int r = ib * TILE + lr, c = kb * TILE + lc;
-#pragma unroll
-        for (int q = 0; q < 16; ++q)
-            rowt[lr * TILE + lc + q] =
-                (r < N && c + q < N) ? D[(size_t)r * N + c + q] : INFINITY;
-        r = kb * TILE + lr;
-        c = jb * TILE + lc;
-#pragma unroll
-        for (int q = 0; q < 16; ++q)
-            colt[lr * TILE + lc + q] =
-                (r < N && c + q < N) ? D[(size_t)r * N + c + q] : INFINITY;
-    }
-    __syncthreads();
-    const int r0 = ib * TILE + ty * 4;
-    const int c0 = jb * TILE + tx * 4;
-    float4 v[4];
-#pragma unroll
-    for (int q = 0; q < 4; ++q) {
-        v[q] = {INFINITY, INFINITY, INFINITY, INFINITY};
-        if (r0 + q < N)
-#pragma unroll
-            for (int p = 0; p < 4; ++p)
-                if (c0 + p < N)
-                    (&v[q].x)[p] = D[(size_t)(r0 + q) * N + c0 + p];
-    }
-#pragma unroll 4
-    for (int k = 0; k < TILE; ++k) {
-        const float4 ckj =
-            *reinterpret_cast<const float4*>(colt + k * TILE + tx * 4);
-#pragma unroll
-        for (int q = 0; q < 4; ++q) {
-            const float dik = rowt[(ty * 4 + q) * TILE + k];
-            v[q].x = fminf(v[q].x, dik + ckj.x);
-            v[q].y = fminf(v[q].y, dik + ckj.y);
-            v[q].z = fminf(v[q].z, dik + ckj.z);
-            v[q].w = fminf(v[q].w, dik + ckj.w);
-        }
-    }
-#pragma unroll
-    for (int q = 0; q < 4; ++q)
-        if (r0 + q < N)
-#pragma unroll
-            for (int p = 0; p < 4; ++p)
-                if (c0 + p < N)
-                    D[(size_t)(r0 + q) * N + c0 + p] = (&v[q].x)[p];
-}
-
-__global__ void fw_phase3_f32_v1(float* __restrict__ d, int N, int nb, int kb) {
-    __shared__ __attribute__((aligned(16))) float rowt[TILE * TILE];
-    __shared__ __attribute__((aligned(16))) float colt[TILE * TILE];
-    float* D = d + (size_t)blockIdx.z * N * N;
-    int ib = blockIdx.y, jb = blockIdx.x;
-    if (ib >= kb) ib += 1;
-    if (jb >= kb) jb += 1;
-    if (ib >= nb || jb >= nb) return;
-    const int tx = threadIdx.x;              // 0..7 → 4 columns each
-    const int ty = threadIdx.y;              // 0..31
-    // cooperative tile loads (4 cols per thread)
-    {
-        const int c4 = tx * 4;
-        int r = ib * TILE + ty, c = kb * TILE + c4;
-#pragma unroll
-        for (int q = 0; q < 4; ++q)
-            rowt[ty * TILE + c4 + q] =
-                (r < N && c + q < N) ? D[(size_t)r * N + c + q] : INFINITY;
-        r = kb * TILE + ty;
-        c = jb * TILE + c4;
-#pragma unroll
-        for (int q = 0; q < 4; ++q)
-            colt[ty * TILE + c4 + q] =
-                (r < N && c + q < N) ? D[(size_t)r * N + c + q] : INFINITY;
-    }
-    __syncthreads();
-    const int r = ib * TILE + ty;
-    const int c0 = jb * TILE + tx * 4;
-    float4 v = {INFINITY, INFINITY, INFINITY, INFINITY};
-    if (r < N) {
-#pragma unroll
-        for (int q = 0; q < 4; ++q)
-            if (c0 + q < N)
-                (&v.x)[q] = D[(size_t)r * N + c0 + q];
-#pragma unroll
-        for (int k = 0; k < TILE; ++k) {
-            const float dik = rowt[ty * TILE + k];
-            const float4 ckj =
-                *reinterpret_cast<const float4*>(colt + k * TILE + tx * 4);
-            v.x = fminf(v.x, dik + ckj.x);
-            v.y = fminf(v.y, dik + ckj.y);
-            v.z = fminf(v.z, dik + ckj.z);
-            v.w = fminf(v.w, dik + ckj.w);
-        }
-#pragma unroll
-        for (int q = 0; q < 4; ++q)
-            if (c0 + q < N)
-                D[(size_t)r * N + c0 + q] = (&v.x)[q];
-    }
-}
-
 template <typename T>
 __global__ void fw_phase3(T* __restrict__ d, int N, int nb, int kb) {
     __shared__ T rowt[TILE * TILE];   // tile (ib, kb)
