@@ -273,3 +273,22 @@ def test_train_replay_cpu(tmp_path):
         "--log_every", "2", "--model_root", str(tmp_path),
         "--training_set", "RPLC"])
     assert history, "no log records"
+
+
+def test_evaluate_refine_cli(tmp_path):
+    """harness.evaluate --refine runs end to end and reports the GNN
+    summary; refined congestion is never above unrefined."""
+    import json
+    from multihop_offload_amd.harness import evaluate as ev
+    out0 = str(tmp_path / "e0.json")
+    out2 = str(tmp_path / "e2.json")
+    common = ["--training_set", "NOPE", "--model_root", str(tmp_path),
+              "--sizes", "20", "--cases-per-size", "4", "--instances", "2",
+              "--T", "300", "--load", "0.5", "--workers", "0",
+              "--seed", "31", "--device", "cpu"]
+    ev.main(common + ["--refine", "0", "--out", out0])
+    ev.main(common + ["--refine", "3", "--out", out2])
+    a = json.load(open(out0))["summary"]["GNN"]
+    b = json.load(open(out2))["summary"]["GNN"]
+    assert b["congest_ratio"] <= a["congest_ratio"]
+    assert b["tau"] <= a["tau"] + 1e-9
