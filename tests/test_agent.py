@@ -272,8 +272,9 @@ def test_log_scalar_jsonl(tmp_path):
 
 def test_chebconv_k3_matches_dense_polynomials(small_case):
     """K=3 layer math vs dense Chebyshev polynomials of the extended
-    adjacency: sum_k T_k(A) X W_k + b with T2 = 2A·T1 - T0 (the GPU path
-    for K>=3 falls back to these torch layers)."""
+    adjacency: sum_k T_k(A) X W_k + b with T2 = 2A·T1 - T0 (this torch
+    recurrence is the CPU oracle the fused generic-K GPU kernels are
+    tested against in tests/test_gpu.py)."""
     from multihop_offload_amd.models.chebconv import ChebConvLayer
     from multihop_offload_amd.queueing import ConflictCSR
 
@@ -310,3 +311,31 @@ def test_learning_decay_schedule(small_case, jobs_for):
         agent._apply_one(zero)
         want = 1e-4 * 0.9 ** ((k) / 100.0)   # lr set BEFORE the k-th apply
         assert np.isclose(agent.optimizer.param_groups[0]["lr"], want)
+
+
+def test_chebconv_k5_matches_dense_polynomials(small_case):
+    """K=5: the full recurrence depth the generic-K kernels unroll
+    (two in-place Chebyshev ping-pong swaps) vs dense polynomials."""
+    from multihop_offload_amd.models.chebconv import ChebConvLayer
+    from multihop_offload_amd.queueing import ConflictCSR
+
+    g = small_case
+    ext = g.ext
+    Ee = ext.num_edges_ext
+    support = ConflictCSR(ext.ext_indptr, ext.ext_indices, device="cpu")
+    layer = ChebConvLayer(4, 6, K=5, dtype=torch.float64,
+                          gen=torch.Generator().manual_seed(2))
+    x = torch.randn(Ee, 4, dtype=torch.float64,
+                    generator=torch.Generator().manual_seed(3))
+    got = layer(x, support)
+
+    A = np.zeros((Ee, Ee))
+    for r in range(Ee):
+        lo, hi = ext.ext_indptr[r], ext.ext_indptr[r + 1]
+        A[r, ext.ext_indices[lo:hi]] = 1.0
+    At = torch.as_tensor(A, dtype=torch.float64)
+    Ts = [x, At @ x]
+    for k in range(2, 5):
+        Ts.append(2.0 * (At @ Ts[-1]) - Ts[-2])
+    want = sum(Ts[k] @ layer.weight[k] for k in range(5)) + layer.bias
+    assert torch.allclose(got, want, atol=1e-10)
