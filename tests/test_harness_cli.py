@@ -292,3 +292,25 @@ def test_evaluate_refine_cli(tmp_path):
     b = json.load(open(out2))["summary"]["GNN"]
     assert b["congest_ratio"] <= a["congest_ratio"]
     assert b["tau"] <= a["tau"] + 1e-9
+
+
+def test_train_multiseed_selects_winner(tmp_path):
+    """Multi-seed production trainer: runs every seed, evaluates all
+    candidates on one shared fresh set, ships the winner into the target
+    training_set dir with a selection report."""
+    import json
+    from multihop_offload_amd.harness import train_multiseed
+    results = train_multiseed.main([
+        "--seeds", "5,6", "--training_set", "MS",
+        "--model_root", str(tmp_path), "--select_cases", "3",
+        "--select_instances", "1",
+        # pass-through trainer args
+        "--steps", "4", "--batch", "8", "--sizes", "20", "--distinct", "3",
+        "--workers", "0", "--device", "cpu", "--guard_every", "0",
+        "--eval_every", "2", "--eval_rounds", "1",
+        "--save_every", "100", "--log_every", "100"])
+    assert len(results) == 2
+    rep = json.load(open(tmp_path / "multiseed_MS.json"))
+    assert rep["winner"]["metric"] == min(r["metric"] for r in results)
+    assert (tmp_path / "model_ChebConv_MS_a5_c5_ACO_agent"
+            / "cp-9999.ckpt.npz").exists()
