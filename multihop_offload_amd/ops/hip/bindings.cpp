@@ -47,6 +47,10 @@ std::vector<torch::Tensor> cheb_bwd_hip(
     torch::Tensor W, torch::Tensor ext_indptr, torch::Tensor ext_base,
     torch::Tensor ext_cols, long max_nnz);
 
+std::vector<torch::Tensor> cheb_bwd_hip_mask(
+    torch::Tensor dlam, torch::Tensor acts, torch::Tensor t1s,
+    torch::Tensor W, torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor ext_cols, long max_nnz, long stage_mask);
 std::vector<torch::Tensor> cheb_kn_fwd_hip(
     torch::Tensor x, torch::Tensor W, torch::Tensor bias,
     torch::Tensor ext_indptr, torch::Tensor ext_base, torch::Tensor ext_cols,
@@ -78,6 +82,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("actor_head_bwd", &actor_head_bwd_hip);
     m.def("cheb_fwd", &cheb_fwd_hip);
     m.def("cheb_bwd", &cheb_bwd_hip);
+    m.def("cheb_bwd_ablate", &cheb_bwd_hip_mask);
     m.def("cheb_kn_fwd", &cheb_kn_fwd_hip);
     m.def("cheb_kn_bwd", &cheb_kn_bwd_hip);
     m.def("fused_adam", &fused_adam_hip);

@@ -377,6 +377,8 @@ __global__ void cheb_fwd_kernel(
 // db (B,L,32); dX not needed at layer 0.
 // LDS: Ab | Db | Tb | Wl | scratch
 // ---------------------------------------------------------------------------
+// stage_mask (timing ablation only — outputs are wrong unless 0xF):
+// bit0 act-mask+db, bit1 load_acts+wgrads, bit2 dx gemm, bit3 spmv
 __global__ void cheb_bwd_kernel(
     const float* __restrict__ dlam,      // (B,Ee)
     const float* __restrict__ acts,      // (B,L+1,Ee,32)
@@ -388,7 +390,7 @@ __global__ void cheb_bwd_kernel(
     float* __restrict__ dW,              // (B,L,K,32,32) out (prezeroed)
     float* __restrict__ db,              // (B,L,32) out (prezeroed)
     int B, int Ee, int L, int K, int rows_pad, int max_nnz,
-    int stage_csr) {
+    int stage_csr, int stage_mask) {
     extern __shared__ __attribute__((aligned(16))) char smem_raw[];
     float* Ab = reinterpret_cast<float*>(smem_raw);   // X_l, later dX
     float* Db = Ab + (size_t)rows_pad * STRIDE;       // current delta
@@ -424,6 +426,7 @@ __global__ void cheb_bwd_kernel(
         const bool last = (l == L - 1);
         // activation mask from stored post-act X_{l+1} (coalesced float4)
         const float slope = last ? 0.f : 0.2f;
+        if (stage_mask & 1)
         for (int t = tid; t < Ee * (F / 4); t += nt) {
             const int r = t >> 3;
             const int c = (t & 7) * 4;
@@ -440,34 +443,38 @@ __global__ void cheb_bwd_kernel(
         // ~400-cycle loads; and folding db into the mask sweep above was
         // measured +13% — atomic flush every column-group change.
         // profiles/r02_notes.md: measure, don't guess.)
-        load_acts(Ab, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
+        if (stage_mask & 2)
+            load_acts(Ab, actsb + (size_t)l * Ee * F, Ee, rows_pad, tid, nt);
         for (int i = tid; i < K * F * F; i += nt)
             Wl[i] = W[((size_t)l * K) * F * F + i];
         __syncthreads();
 
         // db[j] = sum_r Db[r][j] — (j, row-chunk) threads, db prezeroed
-        {
+        if (stage_mask & 1) {
             const int nchunk = nt / F;
             const int j = tid % F, ch = tid / F;
             float acc = 0.f;
             for (int r = ch; r < Ee; r += nchunk) acc += Db[r * STRIDE + j];
             atomicAdd(&dbb[l * F + j], acc);
         }
-        gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
-        if (K > 1) {
-            __syncthreads();
-            load_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee, rows_pad,
-                      tid, nt);                           // T1 from forward
-            __syncthreads();
-            gemm_wgrad(Tb, Db, dWb + ((size_t)l * K + 1) * F * F, rows_pad,
-                       tid);
+        if (stage_mask & 2) {
+            gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
+            if (K > 1) {
+                __syncthreads();
+                load_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee,
+                          rows_pad, tid, nt);             // T1 from forward
+                __syncthreads();
+                gemm_wgrad(Tb, Db, dWb + ((size_t)l * K + 1) * F * F,
+                           rows_pad, tid);
+            }
         }
         if (l == 0) break;                       // features are leaves
         __syncthreads();
         // U = Db·W1ᵀ (into Tb) and dX = Db·W0ᵀ (into Ab), fused per tile
-        gemm_dx_fused(Db, Ab, Tb, Wl, K, rows_pad, tid);
+        if (stage_mask & 4)
+            gemm_dx_fused(Db, Ab, Tb, Wl, K, rows_pad, tid);
         __syncthreads();
-        if (K > 1) {
+        if ((stage_mask & 8) && K > 1) {
             spmv(Tb, Ab, ipt, cls, Ee, rows_pad, tid, nt, 1); // dX += A·U
             __syncthreads();
         }
@@ -777,10 +784,10 @@ std::vector<torch::Tensor> cheb_kn_bwd_hip(
     return {dW, db};
 }
 
-std::vector<torch::Tensor> cheb_bwd_hip(
+std::vector<torch::Tensor> cheb_bwd_hip_mask(
     torch::Tensor dlam, torch::Tensor acts, torch::Tensor t1s,
     torch::Tensor W, torch::Tensor ext_indptr, torch::Tensor ext_base,
-    torch::Tensor ext_cols, long max_nnz) {
+    torch::Tensor ext_cols, long max_nnz, long stage_mask) {
     const int B = dlam.size(0), Ee = dlam.size(1);
     const int L = W.size(0), K = W.size(1);
     const int rows_pad = round16(Ee);
@@ -800,6 +807,15 @@ std::vector<torch::Tensor> cheb_bwd_hip(
                        W.data_ptr<float>(), ext_indptr.data_ptr<int>(),
                        ext_base.data_ptr<long>(), ext_cols.data_ptr<int>(),
                        dW.data_ptr<float>(), db.data_ptr<float>(),
-                       B, Ee, L, K, rows_pad, (int)max_nnz, stage_csr);
+                       B, Ee, L, K, rows_pad, (int)max_nnz, stage_csr,
+                       (int)stage_mask);
     return {dW, db};
+}
+
+std::vector<torch::Tensor> cheb_bwd_hip(
+    torch::Tensor dlam, torch::Tensor acts, torch::Tensor t1s,
+    torch::Tensor W, torch::Tensor ext_indptr, torch::Tensor ext_base,
+    torch::Tensor ext_cols, long max_nnz) {
+    return cheb_bwd_hip_mask(dlam, acts, t1s, W, ext_indptr, ext_base,
+                             ext_cols, max_nnz, 0xF);
 }
