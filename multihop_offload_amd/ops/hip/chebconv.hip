@@ -458,11 +458,15 @@ __global__ void cheb_bwd_kernel(
             atomicAdd(&dbb[l * F + j], acc);
         }
         if (stage_mask & 2) {
-            gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
-            if (K > 1) {
-                __syncthreads();
+            if (K > 1)
+                // issue the T1 staging loads BEFORE wgrad0: disjoint sets
+                // (reads t1s/global, writes Tb rows vs wgrad0's Ab/Db
+                // reads + dW atomics), so no barrier between them — the
+                // global-load latency hides under the wgrad0 MFMA work
                 load_acts(Tb, t1s + ((size_t)b * L + l) * Ee * F, Ee,
                           rows_pad, tid, nt);             // T1 from forward
+            gemm_wgrad(Ab, Db, dWb + ((size_t)l * K) * F * F, rows_pad, tid);
+            if (K > 1) {
                 __syncthreads();
                 gemm_wgrad(Tb, Db, dWb + ((size_t)l * K + 1) * F * F,
                            rows_pad, tid);
